@@ -1,0 +1,58 @@
+// HTTP registry server — route table + handlers
+// (reference: pkg/registry/route.go:15-51, registry.go:18-271).
+#pragma once
+
+#include <memory>
+#include <string>
+
+#include "modelx/http.hpp"
+#include "modelx/store.hpp"
+
+namespace modelx {
+namespace registry {
+
+struct AuthConfig {
+  // Static bearer tokens (comma-separated via --auth-tokens); empty = open.
+  std::vector<std::string> tokens;
+  // HS256 JWT shared secret (offline OIDC-style verification); empty = off.
+  std::string jwt_hs256_secret;
+};
+
+class Registry {
+ public:
+  Registry(std::shared_ptr<store::RegistryStore> s, AuthConfig auth = {})
+      : store_(std::move(s)), auth_(std::move(auth)) {}
+
+  // the single mux entry point (LoggingFilter wraps this in main)
+  void handle(http::Request& req, http::ResponseWriter& w);
+
+ private:
+  bool authorize(http::Request& req, http::ResponseWriter& w);
+
+  void get_global_index(http::Request&, http::ResponseWriter&);
+  void get_index(http::Request&, http::ResponseWriter&, const std::string& name);
+  void delete_index(http::Request&, http::ResponseWriter&, const std::string& name);
+  void get_manifest(http::Request&, http::ResponseWriter&, const std::string& name,
+                    const std::string& ref);
+  void put_manifest(http::Request&, http::ResponseWriter&, const std::string& name,
+                    const std::string& ref);
+  void delete_manifest(http::Request&, http::ResponseWriter&, const std::string& name,
+                       const std::string& ref);
+  void head_blob(http::Request&, http::ResponseWriter&, const std::string& name,
+                 const std::string& digest);
+  void get_blob(http::Request&, http::ResponseWriter&, const std::string& name,
+                const std::string& digest);
+  void put_blob(http::Request&, http::ResponseWriter&, const std::string& name,
+                const std::string& digest);
+  void blob_location(http::Request&, http::ResponseWriter&, const std::string& name,
+                     const std::string& digest, const std::string& purpose);
+  void garbage_collect(http::Request&, http::ResponseWriter&, const std::string& name);
+
+  std::shared_ptr<store::RegistryStore> store_;
+  AuthConfig auth_;
+};
+
+void response_error(http::ResponseWriter& w, const wire::ErrorInfo& e);
+
+}  // namespace registry
+}  // namespace modelx

@@ -1,0 +1,326 @@
+#include "modelx/registry.hpp"
+
+#include <cstring>
+#include <vector>
+
+namespace modelx {
+namespace registry {
+
+static constexpr int64_t kMaxManifestBytes = 1 << 20;  // helper.go:19 MaxBytesRead
+
+void response_error(http::ResponseWriter& w, const wire::ErrorInfo& e) {
+  w.write_json(e.http_status, e.to_json_body());
+}
+
+static void response_ok(http::ResponseWriter& w, const json::Value& v) {
+  // reference ResponseOK uses json.Encoder → trailing newline
+  w.write_json(200, v.dump() + "\n");
+}
+
+// split path into segments (already url-decoded)
+static std::vector<std::string> segments(const std::string& path) {
+  std::vector<std::string> out;
+  size_t pos = 1;  // skip leading /
+  while (pos <= path.size()) {
+    size_t slash = path.find('/', pos);
+    if (slash == std::string::npos) slash = path.size();
+    if (slash > pos) out.push_back(path.substr(pos, slash - pos));
+    pos = slash + 1;
+  }
+  return out;
+}
+
+// NameRegexp component check (route.go:10): alnum groups joined by [._-]
+static bool valid_name_component(const std::string& s) {
+  if (s.empty()) return false;
+  bool prev_sep = true;
+  for (char c : s) {
+    bool alnum = (c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z') || (c >= '0' && c <= '9');
+    if (alnum) {
+      prev_sep = false;
+    } else if (c == '.' || c == '_' || c == '-') {
+      if (prev_sep) return false;
+      prev_sep = true;
+    } else {
+      return false;
+    }
+  }
+  return !prev_sep;
+}
+
+// ReferenceRegexp (route.go:11)
+static bool valid_reference(const std::string& s) {
+  if (s.empty() || s.size() > 128) return false;
+  char c0 = s[0];
+  bool ok0 = (c0 >= 'a' && c0 <= 'z') || (c0 >= 'A' && c0 <= 'Z') || (c0 >= '0' && c0 <= '9') ||
+             c0 == '_';
+  if (!ok0) return false;
+  for (char c : s) {
+    bool ok = (c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z') || (c >= '0' && c <= '9') ||
+              c == '.' || c == '_' || c == '-';
+    if (!ok) return false;
+  }
+  return true;
+}
+
+bool Registry::authorize(http::Request& req, http::ResponseWriter& w) {
+  if (auth_.tokens.empty() && auth_.jwt_hs256_secret.empty()) return true;
+  std::string token;
+  auto it = req.headers.find("Authorization");
+  if (it != req.headers.end() && it->second.rfind("Bearer ", 0) == 0)
+    token = it->second.substr(7);
+  if (token.empty()) {
+    // `?token=` fallback (reference: pkg/registry/helper.go:69-74)
+    auto q = req.query.find("token");
+    if (q != req.query.end()) token = q->second;
+  }
+  if (token.empty()) {
+    response_error(w, wire::ErrorInfo{401, "UNAUTHORIZED", "missing bearer token", ""});
+    return false;
+  }
+  for (auto& t : auth_.tokens)
+    if (t == token) return true;
+  if (!auth_.jwt_hs256_secret.empty()) {
+    extern bool verify_jwt_hs256(const std::string& token, const std::string& secret,
+                                 std::string* subject);
+    std::string subject;
+    if (verify_jwt_hs256(token, auth_.jwt_hs256_secret, &subject)) return true;
+  }
+  response_error(w, wire::ErrorInfo{401, "UNAUTHORIZED", "invalid token", ""});
+  return false;
+}
+
+void Registry::handle(http::Request& req, http::ResponseWriter& w) {
+  const std::string& m = req.method;
+  if (req.path == "/healthz" && m == "GET") {
+    w.write_all(200, "ok");
+    return;
+  }
+  if (req.path == "/" || req.path.empty()) {
+    if (m == "GET") {
+      if (!authorize(req, w)) return;
+      get_global_index(req, w);
+      return;
+    }
+    w.write_all(405, "method not allowed");
+    return;
+  }
+  auto seg = segments(req.path);
+  // all repository routes need name = <project>/<name> (route.go:10 NameRegexp)
+  if (seg.size() < 3 || !valid_name_component(seg[0]) || !valid_name_component(seg[1])) {
+    response_error(w, wire::ErrorInfo{404, "NAME_INVALID", "invalid repository path", req.path});
+    return;
+  }
+  if (!authorize(req, w)) return;
+  std::string name = seg[0] + "/" + seg[1];
+
+  if (seg.size() == 3 && seg[2] == "garbage-collect" && m == "POST") {
+    garbage_collect(req, w, name);
+    return;
+  }
+  if (seg.size() == 3 && seg[2] == "index") {
+    if (m == "GET") return get_index(req, w, name);
+    if (m == "DELETE") return delete_index(req, w, name);
+    w.write_all(405, "method not allowed");
+    return;
+  }
+  if (seg.size() == 4 && seg[2] == "manifests") {
+    const std::string& ref = seg[3];
+    if (!valid_reference(ref)) {
+      response_error(w, wire::ErrorInfo{404, "NAME_INVALID", "invalid reference", ref});
+      return;
+    }
+    if (m == "GET") return get_manifest(req, w, name, ref);
+    if (m == "PUT") return put_manifest(req, w, name, ref);
+    if (m == "DELETE") return delete_manifest(req, w, name, ref);
+    w.write_all(405, "method not allowed");
+    return;
+  }
+  if (seg.size() == 4 && seg[2] == "blobs") {
+    const std::string& digest = seg[3];
+    if (!wire::digest_valid(digest)) {
+      response_error(w, wire::ErrorInfo{400, "DIGEST_INVALID", "digest invalid: " + digest, ""});
+      return;
+    }
+    if (m == "HEAD") return head_blob(req, w, name, digest);
+    if (m == "GET") return get_blob(req, w, name, digest);
+    if (m == "PUT") return put_blob(req, w, name, digest);
+    w.write_all(405, "method not allowed");
+    return;
+  }
+  if (seg.size() == 6 && seg[2] == "blobs" && seg[4] == "locations" && m == "GET") {
+    const std::string& digest = seg[3];
+    if (!wire::digest_valid(digest)) {
+      response_error(w, wire::ErrorInfo{400, "DIGEST_INVALID", "digest invalid: " + digest, ""});
+      return;
+    }
+    return blob_location(req, w, name, digest, seg[5]);
+  }
+  response_error(w, wire::ErrorInfo{404, "UNKNOWN", "no such route", req.path});
+}
+
+void Registry::get_global_index(http::Request& req, http::ResponseWriter& w) {
+  wire::Index index;
+  std::string search;
+  auto it = req.query.find("search");
+  if (it != req.query.end()) search = it->second;
+  if (!store_->GetGlobalIndex(search, &index)) {
+    response_ok(w, wire::Index{}.to_json());
+    return;
+  }
+  response_ok(w, index.to_json());
+}
+
+void Registry::get_index(http::Request& req, http::ResponseWriter& w, const std::string& name) {
+  wire::Index index;
+  std::string search;
+  auto it = req.query.find("search");
+  if (it != req.query.end()) search = it->second;
+  if (!store_->GetIndex(name, search, &index)) {
+    response_error(w, wire::ErrorInfo{404, "INDEX_UNKNOWN", "index: " + name + " not found", ""});
+    return;
+  }
+  response_ok(w, index.to_json());
+}
+
+void Registry::delete_index(http::Request& req, http::ResponseWriter& w,
+                            const std::string& name) {
+  if (!store_->RemoveIndex(name)) {
+    response_error(w, wire::ErrorInfo{404, "INDEX_UNKNOWN", "index: " + name + " not found", ""});
+    return;
+  }
+  response_ok(w, json::Value("ok"));
+}
+
+void Registry::get_manifest(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                            const std::string& ref) {
+  wire::Manifest manifest;
+  if (!store_->GetManifest(name, ref, &manifest)) {
+    response_error(w,
+                   wire::ErrorInfo{404, "MANIFEST_UNKNOWN", "manifest: " + ref + " not found", ""});
+    return;
+  }
+  response_ok(w, manifest.to_json());
+}
+
+void Registry::put_manifest(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                            const std::string& ref) {
+  if (req.content_length > kMaxManifestBytes) {
+    response_error(w, wire::ErrorInfo{413, "MANIFEST_INVALID", "manifest too large", ""});
+    return;
+  }
+  std::string body;
+  try {
+    body = req.read_body_all(kMaxManifestBytes);
+  } catch (const std::exception& e) {
+    response_error(w, wire::ErrorInfo{400, "MANIFEST_INVALID", e.what(), ""});
+    return;
+  }
+  wire::Manifest manifest;
+  try {
+    manifest = wire::Manifest::from_json(json::parse(body));
+  } catch (const std::exception& e) {
+    response_error(w, wire::ErrorInfo{400, "MANIFEST_INVALID", e.what(), ""});
+    return;
+  }
+  std::string content_type;
+  auto it = req.headers.find("Content-Type");
+  if (it != req.headers.end()) content_type = it->second;
+  std::string err;
+  if (!store_->PutManifest(name, ref, content_type, manifest, &err)) {
+    response_error(w, wire::ErrorInfo{500, "INTERNAL", err, ""});
+    return;
+  }
+  w.write_all(201, "");  // registry.go:106 StatusCreated
+}
+
+void Registry::delete_manifest(http::Request& req, http::ResponseWriter& w,
+                               const std::string& name, const std::string& ref) {
+  if (!store_->ExistsManifest(name, ref)) {
+    response_error(w,
+                   wire::ErrorInfo{404, "MANIFEST_UNKNOWN", "manifest: " + ref + " not found", ""});
+    return;
+  }
+  if (!store_->DeleteManifest(name, ref)) {
+    response_error(w, wire::ErrorInfo{500, "INTERNAL", "delete failed", ""});
+    return;
+  }
+  w.write_all(202, "");  // registry.go:119 StatusAccepted
+}
+
+void Registry::head_blob(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                         const std::string& digest) {
+  if (store_->ExistsBlob(name, digest))
+    w.write_all(200, "");
+  else
+    w.write_all(404, "");
+}
+
+void Registry::get_blob(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                        const std::string& digest) {
+  store::FileMeta meta;
+  auto reader = store_->GetBlob(name, digest, &meta);
+  if (!reader) {
+    response_error(w, wire::ErrorInfo{404, "BLOB_UNKNOWN", "blob: " + digest + " not found", ""});
+    return;
+  }
+  w.set_header("Content-Type",
+               meta.content_type.empty() ? "application/octet-stream" : meta.content_type);
+  w.begin(200, meta.size);
+  int fd = reader->sendfile_fd();
+  if (fd >= 0) {
+    w.sendfile(fd, 0, meta.size);
+    return;
+  }
+  std::vector<char> buf(1 << 20);
+  while (true) {
+    ssize_t r = reader->read(buf.data(), buf.size());
+    if (r <= 0) break;
+    w.write(buf.data(), static_cast<size_t>(r));
+  }
+}
+
+void Registry::put_blob(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                        const std::string& digest) {
+  std::string content_type;
+  auto it = req.headers.find("Content-Type");
+  if (it != req.headers.end()) content_type = it->second;
+  if (content_type.empty()) {
+    response_error(w, wire::ErrorInfo{400, "INVALID_PARAMETER", "content type invalid: empty", ""});
+    return;
+  }
+  bool ok = store_->PutBlob(name, digest, content_type, req.content_length,
+                            [&](char* buf, size_t n) { return req.read_body(buf, n); });
+  if (!ok) {
+    response_error(w, wire::ErrorInfo{500, "INTERNAL", "store put blob failed", ""});
+    return;
+  }
+  w.write_all(201, "");  // registry.go:162 StatusCreated
+}
+
+void Registry::blob_location(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                             const std::string& digest, const std::string& purpose) {
+  std::map<std::string, std::string> properties;
+  for (auto& kv : req.query) properties[kv.first] = kv.second;
+  auto loc = store_->GetBlobLocation(name, digest, purpose, properties);
+  if (!loc.supported) {
+    response_error(w, wire::ErrorInfo{501, "UNSUPPORTED", "blob location not supported", ""});
+    return;
+  }
+  json::Object o;
+  o["provider"] = json::Value(loc.provider);
+  o["purpose"] = json::Value(loc.purpose);
+  o["properties"] = loc.properties;
+  response_ok(w, json::Value(std::move(o)));
+}
+
+void Registry::garbage_collect(http::Request& req, http::ResponseWriter& w,
+                               const std::string& name) {
+  int removed = store_->GCBlobs(name);
+  json::Object o;
+  o["blobs"] = json::Value(removed);
+  response_ok(w, json::Value(std::move(o)));
+}
+
+}  // namespace registry
+}  // namespace modelx
