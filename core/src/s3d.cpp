@@ -324,6 +324,7 @@ void handle(http::Request& req, http::ResponseWriter& w) {
     return;
   }
   if (!authorized(req)) {
+    fprintf(stderr, "s3d: auth failed: %s %s\n", req.method.c_str(), req.target.c_str());
     w.write_all(403, "<Error><Code>SignatureDoesNotMatch</Code></Error>", "application/xml");
     return;
   }
