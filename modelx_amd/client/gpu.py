@@ -1,0 +1,223 @@
+"""GPU data plane: S3 → HBM pulls and HBM → S3 pushes on MI355X.
+
+Orchestrates the native pinned-ring engine (modelx_amd._core, built from
+core/src/engine.cpp + core/hip/sha256.hip):
+
+- ``pull_to_gpu``: manifest → presigned URL per blob → parallel ranged GETs
+  into pinned slots → hipMemcpyAsync onto the device buffer → CDNA4 SHA-256
+  chunk kernel verifies the chunked digest. Returns ``dict[name,
+  torch.Tensor]`` (uint8) resident in HBM.
+- ``push_from_gpu``: chunk-digest the device buffers on GPU, presigned
+  (multi)part upload streamed D2H through the pinned ring, manifest PUT last.
+- multi-GPU fan-out (``broadcast_pull``): rank 0 pulls, RCCL broadcast over
+  xGMI per pipeline chunk via torch.distributed — see ``fanout.py``.
+
+The native extension is REQUIRED on a GPU box: if HIP devices are visible and
+the extension is missing, this module raises instead of silently falling back
+to a CPU path.
+"""
+from __future__ import annotations
+
+from datetime import datetime, timezone
+from typing import Dict, List, Optional, Tuple
+
+from ..wire import digest as dg
+from ..wire import errors as er
+from ..wire import types
+from .registry import RegistryClient
+
+DEFAULT_SLOT_BYTES = 64 << 20
+DEFAULT_NUM_SLOTS = 10
+DEFAULT_NUM_CONNS = 8
+DEFAULT_PART_BYTES = 256 << 20  # push part granularity (multipart)
+
+
+def _core():
+    try:
+        from modelx_amd import _core as core
+    except ImportError as e:  # pragma: no cover
+        raise er.ModelxError(
+            er.ErrCode.INTERNAL,
+            "modelx_amd._core native extension not built "
+            "(python setup_ext.py build_ext --inplace)",
+        ) from e
+    return core
+
+
+def _signed_headers(part: dict) -> Dict[str, str]:
+    out = {}
+    for k, v in (part.get("signedHeader") or {}).items():
+        out[k] = ",".join(v) if isinstance(v, list) else str(v)
+    return out
+
+
+class GpuClient:
+    """Per-device transfer client. One GpuEngine (pinned ring + streams +
+    hash stream) per instance."""
+
+    def __init__(self, registry: str, authorization: str = "", device: int = 0,
+                 num_slots: int = DEFAULT_NUM_SLOTS, slot_bytes: int = DEFAULT_SLOT_BYTES,
+                 num_conns: int = DEFAULT_NUM_CONNS):
+        core = _core()
+        if not core.hip_available():
+            raise er.ModelxError(er.ErrCode.INTERNAL, "no HIP device visible")
+        self.remote = RegistryClient(registry, authorization)
+        self.device = device
+        self.num_conns = num_conns
+        self.engine = core.GpuEngine(device=device, num_slots=num_slots,
+                                     slot_bytes=slot_bytes, num_streams=4)
+        self.last_stats: List[dict] = []
+
+    # ----------------------------------------------------------- helpers --
+
+    def _verify_device_digest(self, ptr: int, size: int, desc: types.Descriptor) -> None:
+        """GPU chunk-digest the landed buffer and compare against the
+        descriptor's chunked digest (or its chunk annotation)."""
+        expect = None
+        algo_cs = dg.algo_chunk_size(desc.digest.split(":", 1)[0]) if desc.digest else None
+        if algo_cs:
+            expect, cs = desc.digest, algo_cs
+        else:
+            note = desc.annotations.get(types.ANNOTATION_CHUNK_DIGEST, "")
+            if note:
+                cs = dg.algo_chunk_size(note.split(":", 1)[0]) or dg.DEFAULT_CHUNK_SIZE
+                expect = note
+        if expect is None:
+            # canonical sha256 only: single sequential chain — GPU-hash via
+            # multibuf (1 chain); still verifies, just not at chunk rate
+            digests = self.engine.sha256_multibuf([(ptr, size)])
+            got = "sha256:" + digests[:32].hex()
+            if got != desc.digest:
+                raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
+                                     f"GPU digest mismatch for {desc.name}: {got}")
+            return
+        leaves = self.engine.sha256_chunk_leaves(ptr, size, cs)
+        got = dg.root_from_leaf_bytes(leaves, cs, size)
+        if got != expect:
+            raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
+                                 f"GPU chunk digest mismatch for {desc.name}: {got} != {expect}")
+
+    def _download_url(self, repository: str, desc: types.Descriptor) -> Tuple[str, Dict[str, str]]:
+        loc = self.remote.get_blob_location(repository, desc, "download")
+        if loc is None:
+            raise er.ModelxError(er.ErrCode.UNSUPPORTED,
+                                 "registry has no presigned download location "
+                                 "(GPU pull needs --enable-redirect)")
+        parts = loc.properties.get("parts") or []
+        if not parts:
+            raise er.ModelxError(er.ErrCode.UNKNOWN, "empty location parts")
+        return parts[0]["url"], _signed_headers(parts[0])
+
+    # -------------------------------------------------------------- pull --
+
+    def pull_blob_to_device(self, repository: str, desc: types.Descriptor,
+                            tensor=None, verify: bool = True) -> "torch.Tensor":
+        import torch
+
+        if tensor is None:
+            tensor = torch.empty(desc.size, dtype=torch.uint8, device=f"cuda:{self.device}")
+        assert tensor.numel() >= desc.size
+        url, headers = self._download_url(repository, desc)
+        stats = self.engine.pull_to_device(url, headers, desc.size, tensor.data_ptr(),
+                                           self.num_conns)
+        stats["name"] = desc.name
+        self.last_stats.append(stats)
+        if verify:
+            self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+        return tensor
+
+    def pull_to_gpu(self, repository: str, version: str = "",
+                    verify: bool = True) -> Dict[str, "torch.Tensor"]:
+        """Pull every file blob of a manifest into HBM. Directory (tar.gz)
+        blobs are landed as raw archive bytes under their blob name."""
+        manifest = self.remote.get_manifest(repository, version)
+        out: Dict[str, "torch.Tensor"] = {}
+        for desc in manifest.blobs:
+            if desc.size == 0:
+                continue
+            out[desc.name] = self.pull_blob_to_device(repository, desc, verify=verify)
+        return out
+
+    # -------------------------------------------------------------- push --
+
+    def digest_device_blob(self, ptr: int, size: int,
+                           chunk_size: int = dg.DEFAULT_CHUNK_SIZE) -> Tuple[str, str]:
+        """(chunked_digest, chunk_digest_annotation) of device memory."""
+        leaves = self.engine.sha256_chunk_leaves(ptr, size, chunk_size)
+        root = dg.root_from_leaf_bytes(leaves, chunk_size, size)
+        return root, root
+
+    def push_blob_from_device(self, repository: str, desc: types.Descriptor, ptr: int,
+                              part_bytes: int = DEFAULT_PART_BYTES,
+                              parallel: int = 4) -> None:
+        """Presigned (multi)part upload of device memory. HEAD-dedup first
+        (push.go:169-177 semantics)."""
+        if self.remote.head_blob(repository, desc.digest):
+            return
+        size = desc.size
+        nparts = max(1, (size + part_bytes - 1) // part_bytes)
+        extra = {"part-count": str(nparts), "multipart": "true"} if nparts > 1 else None
+        loc = self.remote.get_blob_location(repository, desc, "upload", extra=extra)
+        if loc is None:
+            raise er.ModelxError(er.ErrCode.UNSUPPORTED,
+                                 "registry has no presigned upload location")
+        parts = loc.properties.get("parts") or []
+        ranges = []
+        base = size // len(parts)
+        off = 0
+        for i in range(len(parts)):
+            ln = base if i < len(parts) - 1 else size - off
+            ranges.append((off, ln))
+            off += ln
+        if len(parts) == 1:
+            p = parts[0]
+            self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
+                                              _signed_headers(p), ptr, size)
+            return
+        from concurrent.futures import ThreadPoolExecutor
+
+        def send(i):
+            p = parts[i]
+            o, ln = ranges[i]
+            self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
+                                              _signed_headers(p), ptr + o, ln)
+
+        with ThreadPoolExecutor(max_workers=parallel) as pool:
+            list(pool.map(send, range(len(parts))))
+
+    def push_from_gpu(self, repository: str, version: str,
+                      tensors: Dict[str, "torch.Tensor"], config_yaml: str = "",
+                      chunk_size: int = dg.DEFAULT_CHUNK_SIZE,
+                      part_bytes: int = DEFAULT_PART_BYTES) -> types.Manifest:
+        """Digest on GPU → presigned multipart upload → manifest PUT last."""
+        manifest = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
+        now = datetime.now(timezone.utc)
+        # config blob (small, CPU)
+        cfg = config_yaml.encode() or b"description: pushed from GPU\n"
+        cfg_digest = dg.sha256_digest(cfg)
+        manifest.config = types.Descriptor(
+            name="modelx.yaml", media_type=types.MEDIA_TYPE_MODEL_CONFIG_YAML,
+            digest=cfg_digest, size=len(cfg), modified=now)
+        if not self.remote.head_blob(repository, cfg_digest):
+            loc = self.remote.get_blob_location(repository, manifest.config, "upload")
+            if loc is not None:
+                import requests
+
+                p = (loc.properties.get("parts") or [{}])[0]
+                requests.request(p.get("method") or "PUT", p["url"],
+                                 headers=_signed_headers(p), data=cfg).raise_for_status()
+            else:
+                self.remote.upload_blob_content(repository, manifest.config, cfg)
+        for name, t in tensors.items():
+            size = t.numel() * t.element_size()
+            root, note = self.digest_device_blob(t.data_ptr(), size, chunk_size)
+            desc = types.Descriptor(
+                name=name, media_type=types.MEDIA_TYPE_MODEL_FILE, digest=root, size=size,
+                modified=now,
+                annotations={types.ANNOTATION_CHUNK_DIGEST: note,
+                             types.ANNOTATION_CHUNK_SIZE: str(chunk_size)})
+            self.push_blob_from_device(repository, desc, t.data_ptr(), part_bytes=part_bytes)
+            manifest.blobs.append(desc)
+        manifest.blobs = types.sort_descriptors_by_name(manifest.blobs)
+        self.remote.put_manifest(repository, version or "latest", manifest)
+        return manifest

@@ -141,6 +141,14 @@ def chunked_digest_file(path: str, chunk_size: int = DEFAULT_CHUNK_SIZE) -> str:
     return f"{chunked_algo_name(chunk_size)}:{h.hexdigest()}"
 
 
+def root_from_leaf_bytes(leaves: bytes, chunk_size: int, total_len: int) -> str:
+    """Chunked digest string from a packed 32 B/leaf array (what the HIP
+    kernel returns)."""
+    n = len(leaves) // 32
+    parts = [leaves[i * 32 : (i + 1) * 32] for i in range(n)] or [hashlib.sha256(b"").digest()]
+    return f"{chunked_algo_name(chunk_size)}:{chunked_root(parts, chunk_size, total_len)}"
+
+
 class StreamingDigester:
     """Incremental digest over a byte stream, computing BOTH the canonical
     sha256 and the chunked digest in one pass (used by the CPU push path;
