@@ -1,0 +1,216 @@
+#!/usr/bin/env python3
+"""modelx_amd flagship benchmark — push+pull GiB/s end-to-end (S3→HBM,
+digest-verified) on 1..8 MI355X (BASELINE.json metric).
+
+One *step* per rank = push one synthetic random blob from HBM to S3 via
+presigned multipart (GPU chunk-digest first) + pull it back into HBM via the
+pinned-ring ranged-GET engine with CDNA4 SHA-256 verification. Per-GPU work
+is fixed as N grows (weak scaling). The S3 store is the bundled modelx-s3d
+(MinIO stand-in) on tmpfs, modelxd coordinates with --enable-redirect; both
+run on this node — the metric measures the client data plane, the store is
+sized to not be the bottleneck.
+
+  python bench.py --gpus 1 --steps 3 --warmup 1            # single GPU
+  torchrun --nproc-per-node 8 bench.py --gpus 8 ...        # driver form
+
+Rank 0 prints exactly one JSON line with the whole-job aggregate.
+"""
+import argparse
+import json
+import os
+import shutil
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "tests"))
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--blob-gib", type=float, default=4.0,
+                   help="per-rank blob size per step (GiB)")
+    p.add_argument("--conns", type=int, default=8, help="ranged-GET connections per rank")
+    p.add_argument("--slot-mib", type=int, default=64)
+    p.add_argument("--slots", type=int, default=10)
+    p.add_argument("--part-mib", type=int, default=256, help="push part size")
+    p.add_argument("--store", default="", help="s3d store root (default: tmpfs)")
+    p.add_argument("--cpu-smoke", action="store_true",
+                   help="CPU-only plumbing run (BASELINE config 1, 10 MiB)")
+    return p.parse_args()
+
+
+def pick_store_root(args):
+    if args.store:
+        return args.store
+    if os.path.isdir("/dev/shm"):
+        return "/dev/shm/modelx-bench"
+    return os.path.join(REPO, "gpurun_out", "bench-store")
+
+
+def cpu_smoke(args):
+    """BASELINE config 1: init + push/pull one 10 MiB blob, CPU only."""
+    import tempfile
+
+    from modelx_amd.client import Client
+    from modelx_amd.config import ModelConfig
+    from util_servers import start_modelxd_s3, start_s3d
+
+    work = tempfile.mkdtemp(prefix="modelx-cpu-bench-")
+    s3d = start_s3d(os.path.join(work, "s3"))
+    mdx = start_modelxd_s3(s3d.url, redirect=True)
+    try:
+        d = os.path.join(work, "model")
+        os.makedirs(d)
+        with open(os.path.join(d, "modelx.yaml"), "w") as f:
+            f.write(ModelConfig(description="bench").to_yaml())
+        with open(os.path.join(d, "blob.bin"), "wb") as f:
+            f.write(os.urandom(10 << 20))
+        c = Client(mdx.url)
+        t0 = time.monotonic()
+        c.push("bench/cfg1", "v1", d, quiet=True)
+        out = os.path.join(work, "out")
+        c.pull("bench/cfg1", "v1", out, quiet=True)
+        dt = time.monotonic() - t0
+        print(json.dumps({
+            "metric": "push+pull GiB/s end-to-end (S3->HBM, digest-verified), 1/2/4/8 MI355X",
+            "value": round((20 / 1024) / dt, 4), "unit": "GiB/s", "n_gpus": 0,
+            "steps": 1, "warmup": 0, "ms_per_step": round(dt * 1e3, 2),
+            "higher_is_better": True, "scaling": "weak", "vs_baseline": None,
+            "dtype": "uint8", "data": "synthetic",
+            "config": {"model": "config1-10MiB-cpu-plumbing", "global_batch": 1,
+                       "seq_len": 0, "parallelism": "cpu"}}))
+    finally:
+        mdx.stop()
+        s3d.stop()
+        shutil.rmtree(work, ignore_errors=True)
+
+
+def main():
+    args = parse_args()
+    if args.cpu_smoke:
+        cpu_smoke(args)
+        return
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    distributed = world > 1
+
+    if distributed:
+        import torch.distributed as dist
+
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="nccl")  # RCCL over xGMI
+    device = local_rank
+    torch.cuda.set_device(device)
+
+    # --- shared local S3 + registry stack (rank 0 owns the processes) ------
+    from util_servers import MODELXD, S3D, ServerProc, wait_http
+
+    master_port = int(os.environ.get("MASTER_PORT", "29500"))
+    s3_port = master_port + 1371
+    mdx_port = master_port + 1372
+    store_root = pick_store_root(args)
+    procs = []
+    if rank == 0:
+        shutil.rmtree(store_root, ignore_errors=True)
+        os.makedirs(os.path.join(store_root, "modelx"), exist_ok=True)
+        if not (os.path.exists(S3D) and os.path.exists(MODELXD)):
+            import subprocess
+
+            subprocess.run(["make", "servers"], cwd=REPO, check=True)
+        procs.append(ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root", store_root,
+                                 "--access-key", "modelx", "--secret-key", "modelx123"], s3_port))
+        wait_http(s3_port)
+        procs.append(ServerProc(
+            [MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
+             f"http://127.0.0.1:{s3_port}", "--s3-bucket", "modelx", "--s3-access-key",
+             "modelx", "--s3-secret-key", "modelx123", "--enable-redirect"], mdx_port))
+        wait_http(mdx_port)
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+
+    from modelx_amd.client.gpu import GpuClient
+    from modelx_amd.client.registry import RegistryClient
+
+    g = GpuClient(f"http://127.0.0.1:{mdx_port}", device=device, num_slots=args.slots,
+                  slot_bytes=args.slot_mib << 20, num_conns=args.conns)
+
+    blob_bytes = int(args.blob_gib * (1 << 30))
+    src = torch.empty(blob_bytes, dtype=torch.uint8, device=f"cuda:{device}")
+    repo = f"bench/rank{rank}"
+
+    def one_step(step_idx: int):
+        # fresh random payload → no HEAD-dedup shortcut; new digest every step
+        src.random_(0, 256)
+        torch.cuda.synchronize(device)
+        g.push_from_gpu(repo, f"s{step_idx}", {"blob.bin": src},
+                        part_bytes=args.part_mib << 20)
+        g.pull_to_gpu(repo, f"s{step_idx}", verify=True)  # GPU digest verify
+        # drop this step's objects so tmpfs doesn't fill across steps
+        g.remote.delete_index(repo)
+
+    def barrier_sync():
+        torch.cuda.synchronize(device)
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        torch.cuda.synchronize(device)
+
+    for w in range(args.warmup):
+        one_step(1000 + w)
+    barrier_sync()
+    t0 = time.monotonic()
+    for k in range(args.steps):
+        one_step(k)
+    barrier_sync()
+    elapsed = time.monotonic() - t0
+
+    # MAX over ranks
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64, device=f"cuda:{device}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank == 0:
+        moved_gib = 2.0 * args.blob_gib * args.steps * world  # push + pull, all ranks
+        print(json.dumps({
+            "metric": "push+pull GiB/s end-to-end (S3->HBM, digest-verified), 1/2/4/8 MI355X",
+            "value": round(moved_gib / elapsed, 3),
+            "unit": "GiB/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "uint8",
+            "data": "synthetic",
+            "config": {"model": f"synthetic-{args.blob_gib:g}GiB-blob",
+                       "global_batch": world, "seq_len": 0,
+                       "parallelism": f"dp{world}-presigned-s3"}}), flush=True)
+        for p in procs:
+            p.stop()
+        shutil.rmtree(store_root, ignore_errors=True)
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,164 @@
+"""GPU numerics + engine tests (run on MI355X via gpurun).
+
+Kernel results are compared against plain CPU references (hashlib) —
+SURVEY.md §4 kernel-unit strategy."""
+import hashlib
+import os
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+from modelx_amd.wire import digest as dg
+
+pytestmark = pytest.mark.gpu
+
+
+def _engine(device=0):
+    from modelx_amd import _core
+
+    assert _core.hip_available(), "native extension loaded but no HIP device"
+    return _core.GpuEngine(device=device, num_slots=8, slot_bytes=32 << 20, num_streams=4)
+
+
+@pytest.fixture(scope="module")
+def engine():
+    return _engine()
+
+
+class TestSha256Kernels:
+    @pytest.mark.parametrize("size", [1, 55, 56, 63, 64, 65, 127, 128, 1000,
+                                      (1 << 20) - 1, 1 << 20, (1 << 20) + 1,
+                                      (5 << 20) + 12345])
+    def test_chunk_leaves_match_hashlib(self, engine, size):
+        data = torch.randint(0, 256, (size,), dtype=torch.uint8)
+        dev = data.cuda()
+        cs = 1 << 20
+        leaves = engine.sha256_chunk_leaves(dev.data_ptr(), size, cs)
+        raw = data.numpy().tobytes()
+        expect = b"".join(
+            hashlib.sha256(raw[o : o + cs]).digest() for o in range(0, size, cs)
+        )
+        assert leaves == expect
+
+    def test_chunk_leaves_small_chunksize(self, engine):
+        # many chunks per block → exercises multi-block grids
+        size = 256 * 1024 + 7
+        cs = 4096
+        data = torch.randint(0, 256, (size,), dtype=torch.uint8)
+        dev = data.cuda()
+        leaves = engine.sha256_chunk_leaves(dev.data_ptr(), size, cs)
+        raw = data.numpy().tobytes()
+        expect = b"".join(hashlib.sha256(raw[o : o + cs]).digest() for o in range(0, size, cs))
+        assert leaves == expect
+
+    def test_root_matches_python_reference(self, engine):
+        size = (3 << 20) + 17
+        cs = 1 << 20
+        data = torch.randint(0, 256, (size,), dtype=torch.uint8)
+        dev = data.cuda()
+        leaves = engine.sha256_chunk_leaves(dev.data_ptr(), size, cs)
+        got = dg.root_from_leaf_bytes(leaves, cs, size)
+        assert got == dg.chunked_digest(data.numpy().tobytes(), cs)
+
+    def test_multibuf_canonical_sha256(self, engine):
+        bufs = []
+        raws = []
+        for size in [1, 100, 4096, 1 << 18, (1 << 20) + 3]:
+            t = torch.randint(0, 256, (size,), dtype=torch.uint8)
+            raws.append(t.numpy().tobytes())
+            bufs.append(t.cuda())
+        digests = engine.sha256_multibuf([(t.data_ptr(), t.numel()) for t in bufs])
+        for i, raw in enumerate(raws):
+            assert digests[i * 32 : (i + 1) * 32] == hashlib.sha256(raw).digest()
+
+
+class TestEngineTransfers:
+    @pytest.fixture(scope="class")
+    def stack(self, tmp_path_factory):
+        from util_servers import start_modelxd_s3, start_s3d
+
+        s3_root = tmp_path_factory.mktemp("s3-gpu")
+        s3d = start_s3d(str(s3_root))
+        mdx = start_modelxd_s3(s3d.url, redirect=True)
+        yield mdx, s3d
+        mdx.stop()
+        s3d.stop()
+
+    def test_pull_to_device_lands_exact_bytes(self, engine, stack, tmp_path):
+        from modelx_amd.client import Client
+        from modelx_amd.client.extension import ContentSource, get as get_ext
+        from modelx_amd.client.push import parse_manifest
+        from modelx_amd.config import ModelConfig
+
+        mdx, _ = stack
+        d = tmp_path / "m"
+        d.mkdir()
+        (d / "modelx.yaml").write_text(ModelConfig(description="gpu").to_yaml())
+        payload = os.urandom(48 * 1024 * 1024 + 12345)
+        (d / "weights.bin").write_bytes(payload)
+        c = Client(mdx.url)
+        c.push("gpu/pull", "v1", str(d), quiet=True)
+
+        from modelx_amd.client.gpu import GpuClient
+
+        g = GpuClient(mdx.url, device=0)
+        tensors = g.pull_to_gpu("gpu/pull", "v1")  # digest-verified on GPU
+        got = tensors["weights.bin"].cpu().numpy().tobytes()
+        assert got == payload
+
+    def test_pull_detects_corruption(self, engine, stack, tmp_path):
+        from modelx_amd.client import Client
+        from modelx_amd.config import ModelConfig
+        from modelx_amd.wire import errors as er
+        from modelx_amd.wire import paths as pm
+
+        mdx, s3d = stack
+        d = tmp_path / "mc"
+        d.mkdir()
+        (d / "modelx.yaml").write_text(ModelConfig(description="corrupt").to_yaml())
+        (d / "weights.bin").write_bytes(os.urandom(8 * 1024 * 1024))
+        c = Client(mdx.url)
+        manifest = c.push("gpu/corrupt", "v1", str(d), quiet=True)
+        blob = next(b for b in manifest.blobs if b.name == "weights.bin")
+        # flip one byte in the stored object
+        root = s3d.proc.args[s3d.proc.args.index("--root") + 1]
+        obj = os.path.join(root, "modelx", "registry",
+                           pm.blob_digest_path("gpu/corrupt", blob.digest))
+        with open(obj, "r+b") as f:
+            f.seek(4 * 1024 * 1024)
+            b0 = f.read(1)
+            f.seek(4 * 1024 * 1024)
+            f.write(bytes([b0[0] ^ 1]))
+        from modelx_amd.client.gpu import GpuClient
+
+        g = GpuClient(mdx.url, device=0)
+        with pytest.raises(er.ModelxError) as exc:
+            g.pull_to_gpu("gpu/corrupt", "v1")
+        assert exc.value.code == er.ErrCode.DIGEST_INVALID
+
+    def test_push_from_gpu_roundtrip(self, engine, stack):
+        from modelx_amd.client.gpu import GpuClient
+
+        mdx, _ = stack
+        g = GpuClient(mdx.url, device=0)
+        src = torch.randint(0, 256, (24 * 1024 * 1024 + 999,), dtype=torch.uint8,
+                            device="cuda:0")
+        manifest = g.push_from_gpu("gpu/pushed", "v1", {"w.bin": src},
+                                   part_bytes=8 << 20)
+        blob = next(b for b in manifest.blobs if b.name == "w.bin")
+        assert blob.digest.startswith("sha256c1m:")
+        back = g.pull_to_gpu("gpu/pushed", "v1")
+        assert torch.equal(back["w.bin"], src)
+
+    def test_push_dedup_skips_existing(self, engine, stack):
+        from modelx_amd.client.gpu import GpuClient
+
+        mdx, _ = stack
+        g = GpuClient(mdx.url, device=0)
+        src = torch.randint(0, 256, (1 << 20,), dtype=torch.uint8, device="cuda:0")
+        g.push_from_gpu("gpu/dedup", "v1", {"a.bin": src})
+        # same content again: HEAD-dedup path (no error, blob reused)
+        g.push_from_gpu("gpu/dedup", "v2", {"a.bin": src})
+        m = g.remote.get_manifest("gpu/dedup", "v2")
+        assert m.blobs[0].size == 1 << 20
