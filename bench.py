@@ -184,6 +184,19 @@ def main():
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    if os.environ.get("MODELX_BENCH_DEBUG") and rank == 0:
+        import collections
+
+        agg = collections.defaultdict(lambda: [0.0, 0])
+        for s in g.last_stats:
+            ph = s.get("phase", "?")
+            agg[ph][0] += s.get("seconds", 0.0)
+            agg[ph][1] += s.get("bytes", 0)
+        for ph, (secs, byts) in sorted(agg.items()):
+            rate = byts / secs / (1 << 30) if secs else 0
+            print(f"# stage {ph}: {secs:.2f}s {byts / (1 << 30):.2f} GiB "
+                  f"{rate:.2f} GiB/s", file=sys.stderr)
+
     if rank == 0:
         moved_gib = 2.0 * args.blob_gib * args.steps * world  # push + pull, all ranks
         print(json.dumps({

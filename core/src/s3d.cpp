@@ -200,6 +200,8 @@ void list_objects(http::Request& req, http::ResponseWriter& w, const std::string
   for (auto& m : metas) {
     std::string key = dir_part.empty() ? m.name : dir_part + "/" + m.name;
     if (!prefix.empty() && key.compare(0, prefix.size(), prefix) != 0) continue;
+    if (key.find(".parts/") != std::string::npos) continue;  // internal multipart storage
+    if (key.size() > 3 && key.compare(key.size() - 3, 3, ".ct") == 0) continue;
     xml += "<Contents><Key>" + xml_escape(key) + "</Key><Size>" + std::to_string(m.size) +
            "</Size><LastModified>" + m.last_modified + "</LastModified></Contents>";
     if (++count >= 100000) break;
@@ -251,6 +253,42 @@ void list_parts(http::Request& req, http::ResponseWriter& w, const std::string& 
   w.write_all(200, xml, "application/xml");
 }
 
+// Multipart completion is ZERO-COPY: parts are renamed under
+// "<object>.parts/<n>" and a tiny "<object>" manifest file records the
+// layout ("S3DPARTS\n<size part>\n..."). GET/HEAD serve ranges across part
+// files with sendfile. (A byte-concat completion — what a naive stand-in
+// does — costs a full extra write of the object and was ~30% of a bench
+// step; real S3 also keeps parts separate internally.)
+constexpr const char* kPartsMagic = "S3DPARTS\n";
+
+struct PartsManifest {
+  std::vector<std::pair<int64_t, std::string>> parts;  // size, path
+  int64_t total = 0;
+};
+
+bool load_parts_manifest(const std::string& path, PartsManifest* out) {
+  int fd = ::open(path.c_str(), O_RDONLY | O_CLOEXEC);
+  if (fd < 0) return false;
+  char buf[8192];
+  ssize_t r = ::read(fd, buf, sizeof buf - 1);
+  ::close(fd);
+  if (r < static_cast<ssize_t>(strlen(kPartsMagic))) return false;
+  buf[r] = 0;
+  if (strncmp(buf, kPartsMagic, strlen(kPartsMagic)) != 0) return false;
+  const char* p = buf + strlen(kPartsMagic);
+  while (*p) {
+    char name[64];
+    long long sz;
+    int n = 0;
+    if (sscanf(p, "%lld %63s%n", &sz, name, &n) != 2) break;
+    out->parts.emplace_back(sz, path + ".parts/" + name);
+    out->total += sz;
+    p += n;
+    while (*p == '\n' || *p == ' ') p++;
+  }
+  return !out->parts.empty();
+}
+
 void complete_multipart(http::Request& req, http::ResponseWriter& w, const std::string& bucket,
                         const std::string& key, const std::string& upload_id) {
   req.read_body_all(16 << 20);  // part list XML (we trust our ListParts order)
@@ -259,7 +297,7 @@ void complete_multipart(http::Request& req, http::ResponseWriter& w, const std::
   std::vector<std::pair<int, std::string>> parts;
   for (auto& m : fs.List("", false)) {
     if (m.name == ".keyinfo") continue;
-    parts.emplace_back(atoi(m.name.c_str()), dir + "/" + m.name);
+    parts.emplace_back(atoi(m.name.c_str()), m.name);
   }
   if (parts.empty()) {
     w.write_all(400, "<Error><Code>InvalidPart</Code></Error>", "application/xml");
@@ -269,41 +307,36 @@ void complete_multipart(http::Request& req, http::ResponseWriter& w, const std::
   std::string dest = obj_path(bucket, key);
   mkdirs_for(dest);
   if (parts.size() == 1) {
-    if (::rename(parts[0].second.c_str(), dest.c_str()) != 0) {
+    if (::rename((dir + "/" + parts[0].second).c_str(), dest.c_str()) != 0) {
       w.write_all(500, "<Error><Code>InternalError</Code></Error>", "application/xml");
       return;
     }
   } else {
-    std::string tmp = dest + ".tmp" + std::to_string(getpid());
-    int out = ::open(tmp.c_str(), O_WRONLY | O_CREAT | O_TRUNC | O_CLOEXEC, 0644);
-    if (out < 0) {
-      w.write_all(500, "<Error><Code>InternalError</Code></Error>", "application/xml");
-      return;
-    }
+    std::string pdir = dest + ".parts";
+    store::LocalFSProvider(g_cfg.root).Remove(
+        pdir.substr(g_cfg.root.size() + 1), true);  // stale parts from a prior object
+    mkdirs_for(pdir + "/x");
+    std::string manifest = kPartsMagic;
     bool ok = true;
     for (auto& p : parts) {
-      int in = ::open(p.second.c_str(), O_RDONLY | O_CLOEXEC);
-      if (in < 0) {
+      struct stat st;
+      std::string src = dir + "/" + p.second;
+      if (::stat(src.c_str(), &st) != 0 ||
+          ::rename(src.c_str(), (pdir + "/" + p.second).c_str()) != 0) {
         ok = false;
         break;
       }
-      struct stat st;
-      fstat(in, &st);
-      off_t off = 0;
-      int64_t left = st.st_size;
-      while (left > 0) {
-        // kernel-side copy: page cache → page cache, no user-space bounce
-        ssize_t c = ::copy_file_range(in, &off, out, nullptr, static_cast<size_t>(left), 0);
-        if (c <= 0) {
-          ok = false;
-          break;
-        }
-        left -= c;
-      }
-      ::close(in);
-      if (!ok) break;
+      manifest += std::to_string((long long)st.st_size) + " " + p.second + "\n";
     }
-    ::close(out);
+    std::string tmp = dest + ".tmp" + std::to_string(getpid());
+    int out = ok ? ::open(tmp.c_str(), O_WRONLY | O_CREAT | O_TRUNC | O_CLOEXEC, 0644) : -1;
+    if (out >= 0) {
+      ok = ::write(out, manifest.data(), manifest.size()) ==
+           static_cast<ssize_t>(manifest.size());
+      ::close(out);
+    } else {
+      ok = false;
+    }
     if (!ok || ::rename(tmp.c_str(), dest.c_str()) != 0) {
       ::unlink(tmp.c_str());
       w.write_all(500, "<Error><Code>InternalError</Code></Error>", "application/xml");
@@ -422,6 +455,9 @@ void handle(http::Request& req, http::ResponseWriter& w) {
       w.write_all(404, "<Error><Code>NoSuchKey</Code></Error>", "application/xml");
       return;
     }
+    PartsManifest pm;
+    bool is_parts = st.st_size < 8192 && load_parts_manifest(path, &pm);
+    int64_t object_size = is_parts ? pm.total : st.st_size;
     std::string ctype = "application/octet-stream";
     {
       int fd = ::open((path + ".ct").c_str(), O_RDONLY | O_CLOEXEC);
@@ -432,6 +468,7 @@ void handle(http::Request& req, http::ResponseWriter& w) {
         if (r > 0) ctype.assign(buf, static_cast<size_t>(r));
       }
     }
+    st.st_size = object_size;
     int64_t start = 0, length = st.st_size;
     int status = 200;
     auto rit = req.headers.find("Range");
@@ -469,16 +506,46 @@ void handle(http::Request& req, http::ResponseWriter& w) {
     w.set_header("Accept-Ranges", "bytes");
     w.begin(status, length);
     if (req.method == "GET") {
-      int fd = ::open(path.c_str(), O_RDONLY | O_CLOEXEC);
-      if (fd >= 0) {
-        w.sendfile(fd, start, length);
-        ::close(fd);
+      if (!is_parts) {
+        int fd = ::open(path.c_str(), O_RDONLY | O_CLOEXEC);
+        if (fd >= 0) {
+          w.sendfile(fd, start, length);
+          ::close(fd);
+        }
+      } else {
+        // stitch the range across part files (zero-copy sendfile per part)
+        int64_t pos = 0, remaining = length, cursor = start;
+        for (auto& part : pm.parts) {
+          if (remaining <= 0) break;
+          int64_t psize = part.first;
+          if (cursor >= pos + psize) {
+            pos += psize;
+            continue;
+          }
+          int64_t in_off = cursor - pos;
+          int64_t take = std::min(psize - in_off, remaining);
+          int fd = ::open(part.second.c_str(), O_RDONLY | O_CLOEXEC);
+          if (fd < 0) break;
+          bool ok = w.sendfile(fd, in_off, take);
+          ::close(fd);
+          if (!ok) break;
+          cursor += take;
+          remaining -= take;
+          pos += psize;
+        }
       }
     }
     return;
   }
 
   if (req.method == "DELETE") {
+    struct stat st;
+    if (::stat(path.c_str(), &st) == 0 && st.st_size < 8192) {
+      PartsManifest pm;
+      if (load_parts_manifest(path, &pm))
+        store::LocalFSProvider(g_cfg.root)
+            .Remove(path.substr(g_cfg.root.size() + 1) + ".parts", true);
+    }
     ::unlink((path + ".ct").c_str());
     ::unlink(path.c_str());
     w.write_all(204, "");

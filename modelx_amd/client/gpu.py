@@ -117,13 +117,19 @@ class GpuClient:
         if tensor is None:
             tensor = torch.empty(desc.size, dtype=torch.uint8, device=f"cuda:{self.device}")
         assert tensor.numel() >= desc.size
+        import time
+
         url, headers = self._download_url(repository, desc)
         stats = self.engine.pull_to_device(url, headers, desc.size, tensor.data_ptr(),
                                            self.num_conns)
         stats["name"] = desc.name
+        stats["phase"] = "pull-transfer"
         self.last_stats.append(stats)
         if verify:
+            t0 = time.monotonic()
             self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+            self.last_stats.append({"phase": "pull-verify", "bytes": desc.size,
+                                    "seconds": time.monotonic() - t0})
         return tensor
 
     def pull_to_gpu(self, repository: str, version: str = "",
@@ -143,8 +149,13 @@ class GpuClient:
     def digest_device_blob(self, ptr: int, size: int,
                            chunk_size: int = dg.DEFAULT_CHUNK_SIZE) -> Tuple[str, str]:
         """(chunked_digest, chunk_digest_annotation) of device memory."""
+        import time
+
+        t0 = time.monotonic()
         leaves = self.engine.sha256_chunk_leaves(ptr, size, chunk_size)
         root = dg.root_from_leaf_bytes(leaves, chunk_size, size)
+        self.last_stats.append({"phase": "gpu-digest", "bytes": size,
+                                "seconds": time.monotonic() - t0})
         return root, root
 
     def push_blob_from_device(self, repository: str, desc: types.Descriptor, ptr: int,
@@ -152,6 +163,9 @@ class GpuClient:
                               parallel: int = 4) -> None:
         """Presigned (multi)part upload of device memory. HEAD-dedup first
         (push.go:169-177 semantics)."""
+        import time
+
+        t0 = time.monotonic()
         if self.remote.head_blob(repository, desc.digest):
             return
         size = desc.size
@@ -173,17 +187,19 @@ class GpuClient:
             p = parts[0]
             self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
                                               _signed_headers(p), ptr, size)
-            return
-        from concurrent.futures import ThreadPoolExecutor
+        else:
+            from concurrent.futures import ThreadPoolExecutor
 
-        def send(i):
-            p = parts[i]
-            o, ln = ranges[i]
-            self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
-                                              _signed_headers(p), ptr + o, ln)
+            def send(i):
+                p = parts[i]
+                o, ln = ranges[i]
+                self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
+                                                  _signed_headers(p), ptr + o, ln)
 
-        with ThreadPoolExecutor(max_workers=parallel) as pool:
-            list(pool.map(send, range(len(parts))))
+            with ThreadPoolExecutor(max_workers=parallel) as pool:
+                list(pool.map(send, range(len(parts))))
+        self.last_stats.append({"phase": "push-upload", "bytes": size,
+                                "seconds": time.monotonic() - t0, "parts": len(parts)})
 
     def push_from_gpu(self, repository: str, version: str,
                       tensors: Dict[str, "torch.Tensor"], config_yaml: str = "",
