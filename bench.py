@@ -149,15 +149,23 @@ def main():
     src = torch.empty(blob_bytes, dtype=torch.uint8, device=f"cuda:{device}")
     repo = f"bench/rank{rank}"
 
+    substeps = []
+
     def one_step(step_idx: int):
         # fresh random payload → no HEAD-dedup shortcut; new digest every step
+        t0 = time.monotonic()
         src.random_(0, 256)
         torch.cuda.synchronize(device)
+        t1 = time.monotonic()
         g.push_from_gpu(repo, f"s{step_idx}", {"blob.bin": src},
                         part_bytes=args.part_mib << 20)
+        t2 = time.monotonic()
         g.pull_to_gpu(repo, f"s{step_idx}", verify=True)  # GPU digest verify
+        t3 = time.monotonic()
         # drop this step's objects so tmpfs doesn't fill across steps
         g.remote.delete_index(repo)
+        t4 = time.monotonic()
+        substeps.append((t1 - t0, t2 - t1, t3 - t2, t4 - t3))
 
     def barrier_sync():
         torch.cuda.synchronize(device)
@@ -185,6 +193,9 @@ def main():
         elapsed = float(t.item())
 
     if os.environ.get("MODELX_BENCH_DEBUG") and rank == 0:
+        for i, (tg, tp, tl, td) in enumerate(substeps):
+            print(f"# step {i}: gen={tg * 1e3:.0f}ms push={tp * 1e3:.0f}ms "
+                  f"pull={tl * 1e3:.0f}ms delete={td * 1e3:.0f}ms", file=sys.stderr)
         import collections
 
         agg = collections.defaultdict(lambda: [0.0, 0])
