@@ -15,7 +15,12 @@
 #include <unistd.h>
 
 #include <algorithm>
+#include <atomic>
+#include <condition_variable>
 #include <cstring>
+#include <deque>
+#include <mutex>
+#include <thread>
 #include <ctime>
 #include <string>
 
@@ -176,6 +181,53 @@ bool authorized(http::Request& req) {
   std::string expect = sigv4::sign_authorization(rts, cred, amz_date);
   size_t esig = expect.find("Signature=");
   return esig != std::string::npos && expect.substr(esig + 10) == given_sig;
+}
+
+// ---- deferred deletion -----------------------------------------------------
+std::mutex g_trash_mu;
+std::condition_variable g_trash_cv;
+std::deque<std::string> g_trash_queue;
+std::atomic<uint64_t> g_trash_seq{0};
+
+void defer_delete(const std::string& path) {
+  std::string trash_dir = g_cfg.root + "/.trash";
+  mkdirs_for(trash_dir + "/x");
+  uint64_t seq = g_trash_seq.fetch_add(1);
+  ::unlink((path + ".ct").c_str());
+  bool queued = false;
+  std::string t1 = trash_dir + "/" + std::to_string(getpid()) + "-" + std::to_string(seq);
+  if (::rename(path.c_str(), t1.c_str()) == 0) {
+    queued = true;
+  }
+  std::string t2 = t1 + ".parts";
+  if (::rename((path + ".parts").c_str(), t2.c_str()) == 0) queued = true;
+  if (queued) {
+    {
+      std::lock_guard<std::mutex> lk(g_trash_mu);
+      g_trash_queue.push_back(t1);
+      g_trash_queue.push_back(t2);
+    }
+    g_trash_cv.notify_one();
+  }
+}
+
+void trash_collector() {
+  while (true) {
+    std::string victim;
+    {
+      std::unique_lock<std::mutex> lk(g_trash_mu);
+      g_trash_cv.wait(lk, [] { return !g_trash_queue.empty(); });
+      victim = g_trash_queue.front();
+      g_trash_queue.pop_front();
+    }
+    struct stat st;
+    if (::lstat(victim.c_str(), &st) == 0) {
+      if (S_ISDIR(st.st_mode))
+        store::LocalFSProvider(g_cfg.root).Remove(victim.substr(g_cfg.root.size() + 1), true);
+      else
+        ::unlink(victim.c_str());
+    }
+  }
 }
 
 void list_objects(http::Request& req, http::ResponseWriter& w, const std::string& bucket) {
@@ -539,15 +591,10 @@ void handle(http::Request& req, http::ResponseWriter& w) {
   }
 
   if (req.method == "DELETE") {
-    struct stat st;
-    if (::stat(path.c_str(), &st) == 0 && st.st_size < 8192) {
-      PartsManifest pm;
-      if (load_parts_manifest(path, &pm))
-        store::LocalFSProvider(g_cfg.root)
-            .Remove(path.substr(g_cfg.root.size() + 1) + ".parts", true);
-    }
-    ::unlink((path + ".ct").c_str());
-    ::unlink(path.c_str());
+    // deferred deletion: rename into .trash (fast) and let the collector
+    // thread unlink — freeing GiBs of tmpfs pages synchronously would stall
+    // the request path for hundreds of ms
+    defer_delete(path);
     w.write_all(204, "");
     return;
   }
@@ -576,6 +623,7 @@ int main(int argc, char** argv) {
   }
   signal(SIGPIPE, SIG_IGN);
   mkdirs_for(g_cfg.root + "/.");
+  std::thread(trash_collector).detach();
   http::Server server(listen, handle);
   int port = server.start();
   printf("modelx-s3d listening on port %d root=%s\n", port, g_cfg.root.c_str());
