@@ -134,6 +134,8 @@ class ClientConn {
   bool read_response_head(int* status, Headers* headers);
   // reads up to n body bytes; 0 = body complete; <0 = error
   ssize_t read_body(char* buf, size_t n);
+  // reads exactly n body bytes (MSG_WAITALL bulk recv); false on error/EOF
+  bool read_body_exact(char* buf, size_t n);
   void close_fd();
 
   const std::string& host() const { return host_; }

@@ -290,21 +290,25 @@ class GpuEngine {
     for (auto& kv : headers) h[kv.first] = kv.second;
     if (!conn.send_request(method, u.target(), h, static_cast<int64_t>(size)))
       throw std::runtime_error("push: send_request failed");
-    // double-buffered D2H → send
+    // double-buffered D2H → send; streams chosen round-robin so concurrent
+    // part uploads don't serialize on one stream
+    hipStream_t st_a = streams_[push_rr_.fetch_add(1) % streams_.size()];
+    hipStream_t st_b = streams_[push_rr_.fetch_add(1) % streams_.size()];
     Slot* cur = acquire_slot();
     Slot* nxt = acquire_slot();
     uint64_t off = 0;
     uint64_t cur_len = std::min<uint64_t>(slot_bytes_, size);
     HIP_CHECK(hipMemcpyAsync(cur->host, reinterpret_cast<char*>(src_ptr), cur_len,
-                             hipMemcpyDeviceToHost, streams_[0]));
-    HIP_CHECK(hipEventRecord(cur->event, streams_[0]));
+                             hipMemcpyDeviceToHost, st_a));
+    HIP_CHECK(hipEventRecord(cur->event, st_a));
     while (off < size) {
       uint64_t next_off = off + cur_len;
       uint64_t next_len = next_off < size ? std::min<uint64_t>(slot_bytes_, size - next_off) : 0;
       if (next_len) {
         HIP_CHECK(hipMemcpyAsync(nxt->host, reinterpret_cast<char*>(src_ptr) + next_off, next_len,
-                                 hipMemcpyDeviceToHost, streams_[1 % streams_.size()]));
-        HIP_CHECK(hipEventRecord(nxt->event, streams_[1 % streams_.size()]));
+                                 hipMemcpyDeviceToHost, st_b));
+        HIP_CHECK(hipEventRecord(nxt->event, st_b));
+        std::swap(st_a, st_b);
       }
       HIP_CHECK(hipEventSynchronize(cur->event));
       if (!conn.send_body(cur->host, cur_len)) {
@@ -374,18 +378,13 @@ class GpuEngine {
       conn.close_fd();
       return false;
     }
-    uint64_t got = 0;
-    while (got < r.length) {
-      ssize_t n = conn.read_body(dst + got, r.length - got);
-      if (n <= 0) return false;
-      got += static_cast<uint64_t>(n);
-    }
-    // drain any extra (status 200 whole-body case shouldn't happen with Range)
-    return got == r.length;
+    // bulk MSG_WAITALL read straight into the pinned slot
+    return conn.read_body_exact(dst, r.length);
   }
 
   int device_;
   size_t slot_bytes_;
+  std::atomic<int> push_rr_{0};
   std::vector<Slot> slots_;
   std::queue<Slot*> free_;
   std::deque<Slot*> pending_;

@@ -610,6 +610,36 @@ ssize_t ClientConn::read_body(char* buf, size_t n) {
   return r;
 }
 
+bool ClientConn::read_body_exact(char* buf, size_t n) {
+  size_t got = 0;
+  // drain buffered bytes first
+  while (got < n && rpos_ < rbuf_.size()) {
+    ssize_t r = read_body(buf + got, n - got);
+    if (r <= 0) return false;
+    got += static_cast<size_t>(r);
+  }
+  while (got < n) {
+    size_t want = n - got;
+    if (body_remaining_ >= 0 && static_cast<int64_t>(want) > body_remaining_)
+      want = static_cast<size_t>(body_remaining_);
+    if (want == 0) return false;
+    ssize_t r;
+    do {
+      r = ::recv(fd_, buf + got, want, MSG_WAITALL);
+    } while (r < 0 && errno == EINTR);
+    if (r <= 0) {
+      close_fd();
+      return false;
+    }
+    got += static_cast<size_t>(r);
+    if (body_remaining_ > 0) {
+      body_remaining_ -= r;
+      if (body_remaining_ == 0 && !keep_alive_) close_fd();
+    }
+  }
+  return true;
+}
+
 bool ClientConn::do_request(const std::string& method, const std::string& target,
                             const Headers& headers, const std::string& body, ClientResponse* out,
                             size_t max_body) {

@@ -27,9 +27,14 @@ from ..wire import types
 from .registry import RegistryClient
 
 DEFAULT_SLOT_BYTES = 64 << 20
-DEFAULT_NUM_SLOTS = 10
+DEFAULT_NUM_SLOTS = 16
 DEFAULT_NUM_CONNS = 8
 DEFAULT_PART_BYTES = 256 << 20  # push part granularity (multipart)
+DEFAULT_PUSH_PARALLEL = 6
+# 128 KiB chunks put 65k+ SHA-256 chains in flight for multi-GiB blobs — the
+# measured sweet spot on MI355X (1017 GiB/s vs 147 GiB/s at 1 MiB chunks,
+# profiles/gpu_micro.md)
+DEFAULT_GPU_CHUNK = 128 << 10
 
 
 def _core():
@@ -147,7 +152,7 @@ class GpuClient:
     # -------------------------------------------------------------- push --
 
     def digest_device_blob(self, ptr: int, size: int,
-                           chunk_size: int = dg.DEFAULT_CHUNK_SIZE) -> Tuple[str, str]:
+                           chunk_size: int = DEFAULT_GPU_CHUNK) -> Tuple[str, str]:
         """(chunked_digest, chunk_digest_annotation) of device memory."""
         import time
 
@@ -160,7 +165,7 @@ class GpuClient:
 
     def push_blob_from_device(self, repository: str, desc: types.Descriptor, ptr: int,
                               part_bytes: int = DEFAULT_PART_BYTES,
-                              parallel: int = 4) -> None:
+                              parallel: int = DEFAULT_PUSH_PARALLEL) -> None:
         """Presigned (multi)part upload of device memory. HEAD-dedup first
         (push.go:169-177 semantics)."""
         import time
@@ -203,7 +208,7 @@ class GpuClient:
 
     def push_from_gpu(self, repository: str, version: str,
                       tensors: Dict[str, "torch.Tensor"], config_yaml: str = "",
-                      chunk_size: int = dg.DEFAULT_CHUNK_SIZE,
+                      chunk_size: int = DEFAULT_GPU_CHUNK,
                       part_bytes: int = DEFAULT_PART_BYTES) -> types.Manifest:
         """Digest on GPU → presigned multipart upload → manifest PUT last."""
         manifest = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
