@@ -85,36 +85,28 @@ bool write_stream_to(const std::string& path, http::Request& req, std::string* e
   std::string tmp = path + ".tmp" + std::to_string(getpid());
   int fd = ::open(tmp.c_str(), O_WRONLY | O_CREAT | O_TRUNC | O_CLOEXEC, 0644);
   if (fd < 0) return false;
-  std::vector<char> buf(8 << 20);
+  std::vector<char> buf(4 << 20);
   int64_t total = 0;
   while (true) {
-    // fill the buffer before writing: one big write per ~8 MiB
-    size_t fill = 0;
-    ssize_t r = 1;
-    while (fill < buf.size()) {
-      r = req.read_body(buf.data() + fill, buf.size() - fill);
-      if (r <= 0) break;
-      fill += static_cast<size_t>(r);
-    }
+    ssize_t r = req.read_body(buf.data(), buf.size());
     if (r < 0) {
       ::close(fd);
       ::unlink(tmp.c_str());
       return false;
     }
-    if (fill == 0) break;
-    size_t off = 0;
-    while (off < fill) {
-      ssize_t w = ::write(fd, buf.data() + off, fill - off);
+    if (r == 0) break;
+    ssize_t off = 0;
+    while (off < r) {
+      ssize_t w = ::write(fd, buf.data() + off, static_cast<size_t>(r - off));
       if (w < 0) {
         if (errno == EINTR) continue;
         ::close(fd);
         ::unlink(tmp.c_str());
         return false;
       }
-      off += static_cast<size_t>(w);
+      off += w;
     }
-    total += fill;
-    if (r == 0) break;
+    total += r;
   }
   ::close(fd);
   if (total != req.content_length) {

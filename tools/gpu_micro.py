@@ -43,6 +43,8 @@ def main():
     obj = os.path.join(s3root, "modelx", "big.bin")
     with open(obj, "wb") as f:
         f.truncate(size)  # sparse: measures the transfer path, not disk
+    objr = os.path.join(s3root, "modelx", "real.bin")
+    os.system(f"dd if=/dev/zero of={objr} bs=64M count={size >> 26} status=none")
     port = free_port()
     proc = subprocess.Popen([S3D, "--listen", f"127.0.0.1:{port}", "--root", s3root,
                              "--no-auth"], stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL)
@@ -53,6 +55,12 @@ def main():
         for conns in [4, 8, 16, 24, 32]:
             stats = eng.pull_to_device(url, {}, size, buf.data_ptr(), conns)
             print(f"conns={conns}: {stats['gib_per_s']:.1f} GiB/s")
+
+        print("\n-- pull_to_device REAL pages (8 GiB, 16 conns) --")
+        urlr = f"http://127.0.0.1:{port}/modelx/real.bin"
+        for rep in range(2):
+            stats = eng.pull_to_device(urlr, {}, size, buf.data_ptr(), 16)
+            print(f"real rep{rep}: {stats['gib_per_s']:.1f} GiB/s")
 
         print("\n-- pull rate vs slot size (16 conns) --")
         for slot_mib in [16, 32, 64, 128]:
