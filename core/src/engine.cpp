@@ -106,7 +106,7 @@ class GpuEngine {
   // Returns timing stats.
   py::dict pull_to_device(const std::string& url,
                           const std::map<std::string, std::string>& headers, uint64_t size,
-                          uintptr_t dst_ptr, int num_conns) {
+                          uintptr_t dst_ptr, int num_conns, uint64_t base_offset = 0) {
     py::gil_scoped_release release;
     HIP_CHECK(hipSetDevice(device_));
     double t0 = now_s();
@@ -166,7 +166,7 @@ class GpuEngine {
           bool ok = false;
           for (int attempt = 0; attempt < 3 && !ok; attempt++) {
             double tn = now_s();
-            ok = fetch_range(conn, u, h, r, slot->host);
+            ok = fetch_range(conn, u, h, {r.offset + base_offset, r.length}, slot->host);
             if (ok) {
               net_ns.fetch_add(static_cast<long>((now_s() - tn) * 1e9));
               net_bytes.fetch_add(r.length);
@@ -419,7 +419,8 @@ PYBIND11_MODULE(_core, m) {
       .def(py::init<int, int, size_t, int>(), py::arg("device") = 0, py::arg("num_slots") = 8,
            py::arg("slot_bytes") = (size_t)(64 << 20), py::arg("num_streams") = 4)
       .def("pull_to_device", &GpuEngine::pull_to_device, py::arg("url"), py::arg("headers"),
-           py::arg("size"), py::arg("dst_ptr"), py::arg("num_conns") = 8)
+           py::arg("size"), py::arg("dst_ptr"), py::arg("num_conns") = 8,
+           py::arg("base_offset") = 0)
       .def("sha256_chunk_leaves", &GpuEngine::sha256_chunk_leaves, py::arg("dev_ptr"),
            py::arg("size"), py::arg("chunk_size"))
       .def("sha256_multibuf", &GpuEngine::sha256_multibuf, py::arg("buffers"))
