@@ -1,0 +1,79 @@
+"""GPU fan-out path tests (single-GPU degenerate forms; the 8-GPU scaling
+run is the driver's). Exercises RCCL init (world 1) + ranged base_offset
+pulls used by the pipelined broadcast."""
+import os
+
+import pytest
+
+torch = pytest.importorskip("torch")
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def stack(tmp_path_factory):
+    from util_servers import start_modelxd_s3, start_s3d
+
+    s3_root = tmp_path_factory.mktemp("s3-fan")
+    s3d = start_s3d(str(s3_root))
+    mdx = start_modelxd_s3(s3d.url, redirect=True)
+    yield mdx, s3d
+    mdx.stop()
+    s3d.stop()
+
+
+def test_pull_with_base_offset(stack):
+    from modelx_amd.client.gpu import GpuClient
+
+    mdx, _ = stack
+    g = GpuClient(mdx.url, device=0)
+    src = torch.randint(0, 256, (32 << 20,), dtype=torch.uint8, device="cuda:0")
+    g.push_from_gpu("fan/off", "v1", {"w.bin": src})
+    desc = next(b for b in g.remote.get_manifest("fan/off", "v1").blobs if b.name == "w.bin")
+    url, headers = g._download_url("fan/off", desc)
+    dst = torch.zeros(8 << 20, dtype=torch.uint8, device="cuda:0")
+    g.engine.pull_to_device(url, headers, 8 << 20, dst.data_ptr(), 4, 1 << 20)
+    assert torch.equal(dst, src[1 << 20 : 9 << 20])
+
+
+def test_fanout_broadcast_rccl_world1(stack):
+    import torch.distributed as dist
+
+    from modelx_amd.client.fanout import fanout_pull_broadcast
+    from modelx_amd.client.gpu import GpuClient
+
+    mdx, _ = stack
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29771")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        g = GpuClient(mdx.url, device=0)
+        src = torch.randint(0, 256, (24 << 20,), dtype=torch.uint8, device="cuda:0")
+        g.push_from_gpu("fan/bcast", "v1", {"w.bin": src})
+        out = fanout_pull_broadcast(dist, g, "fan/bcast", "v1", device=0,
+                                    chunk=4 << 20)
+        assert torch.equal(out["w.bin"], src)
+    finally:
+        dist.destroy_process_group()
+
+
+def test_fanout_sharded_world1(stack):
+    import torch.distributed as dist
+
+    from modelx_amd.client.fanout import fanout_pull_sharded
+    from modelx_amd.client.gpu import GpuClient
+
+    mdx, _ = stack
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29772")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        g = GpuClient(mdx.url, device=0)
+        tensors = {f"shard{i}.bin": torch.randint(0, 256, (4 << 20,), dtype=torch.uint8,
+                                                  device="cuda:0") for i in range(3)}
+        g.push_from_gpu("fan/shard", "v1", tensors)
+        out = fanout_pull_sharded(dist, g, "fan/shard", "v1", device=0, replicate=True)
+        for name, t in tensors.items():
+            assert torch.equal(out[name], t)
+    finally:
+        dist.destroy_process_group()
