@@ -2,6 +2,7 @@
 // (reference: pkg/registry/route.go:15-51, registry.go:18-271).
 #pragma once
 
+#include <atomic>
 #include <memory>
 #include <string>
 
@@ -16,6 +17,20 @@ struct AuthConfig {
   std::vector<std::string> tokens;
   // HS256 JWT shared secret (offline OIDC-style verification); empty = off.
   std::string jwt_hs256_secret;
+};
+
+// Prometheus-style counters (reference has no metrics endpoint — SURVEY.md
+// §5 observability gap; this is the /metrics the new framework adds).
+struct Metrics {
+  std::atomic<uint64_t> requests_total{0};
+  std::atomic<uint64_t> blob_bytes_in{0};
+  std::atomic<uint64_t> blob_bytes_out{0};
+  std::atomic<uint64_t> presign_upload_total{0};
+  std::atomic<uint64_t> presign_download_total{0};
+  std::atomic<uint64_t> manifests_put_total{0};
+  std::atomic<uint64_t> gc_blobs_removed_total{0};
+  std::atomic<uint64_t> errors_total{0};
+  std::string render() const;
 };
 
 class Registry {
@@ -50,6 +65,7 @@ class Registry {
 
   std::shared_ptr<store::RegistryStore> store_;
   AuthConfig auth_;
+  Metrics metrics_;
 };
 
 void response_error(http::ResponseWriter& w, const wire::ErrorInfo& e);

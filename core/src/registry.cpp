@@ -12,6 +12,22 @@ void response_error(http::ResponseWriter& w, const wire::ErrorInfo& e) {
   w.write_json(e.http_status, e.to_json_body());
 }
 
+std::string Metrics::render() const {
+  auto line = [](const char* name, uint64_t v) {
+    return "# TYPE " + std::string(name) + " counter\n" + name + " " + std::to_string(v) + "\n";
+  };
+  std::string out;
+  out += line("modelx_requests_total", requests_total.load());
+  out += line("modelx_blob_bytes_in_total", blob_bytes_in.load());
+  out += line("modelx_blob_bytes_out_total", blob_bytes_out.load());
+  out += line("modelx_presign_upload_total", presign_upload_total.load());
+  out += line("modelx_presign_download_total", presign_download_total.load());
+  out += line("modelx_manifests_put_total", manifests_put_total.load());
+  out += line("modelx_gc_blobs_removed_total", gc_blobs_removed_total.load());
+  out += line("modelx_errors_total", errors_total.load());
+  return out;
+}
+
 static void response_ok(http::ResponseWriter& w, const json::Value& v) {
   // reference ResponseOK uses json.Encoder → trailing newline
   w.write_json(200, v.dump() + "\n");
@@ -92,8 +108,13 @@ bool Registry::authorize(http::Request& req, http::ResponseWriter& w) {
 
 void Registry::handle(http::Request& req, http::ResponseWriter& w) {
   const std::string& m = req.method;
+  metrics_.requests_total.fetch_add(1);
   if (req.path == "/healthz" && m == "GET") {
     w.write_all(200, "ok");
+    return;
+  }
+  if (req.path == "/metrics" && m == "GET") {
+    w.write_all(200, metrics_.render(), "text/plain; version=0.0.4");
     return;
   }
   if (req.path == "/" || req.path.empty()) {
@@ -228,9 +249,11 @@ void Registry::put_manifest(http::Request& req, http::ResponseWriter& w, const s
   if (it != req.headers.end()) content_type = it->second;
   std::string err;
   if (!store_->PutManifest(name, ref, content_type, manifest, &err)) {
+    metrics_.errors_total.fetch_add(1);
     response_error(w, wire::ErrorInfo{500, "INTERNAL", err, ""});
     return;
   }
+  metrics_.manifests_put_total.fetch_add(1);
   w.write_all(201, "");  // registry.go:106 StatusCreated
 }
 
@@ -266,6 +289,7 @@ void Registry::get_blob(http::Request& req, http::ResponseWriter& w, const std::
   }
   w.set_header("Content-Type",
                meta.content_type.empty() ? "application/octet-stream" : meta.content_type);
+  metrics_.blob_bytes_out.fetch_add(static_cast<uint64_t>(meta.size));
   w.begin(200, meta.size);
   int fd = reader->sendfile_fd();
   if (fd >= 0) {
@@ -289,6 +313,7 @@ void Registry::put_blob(http::Request& req, http::ResponseWriter& w, const std::
     response_error(w, wire::ErrorInfo{400, "INVALID_PARAMETER", "content type invalid: empty", ""});
     return;
   }
+  metrics_.blob_bytes_in.fetch_add(static_cast<uint64_t>(req.content_length));
   bool ok = store_->PutBlob(name, digest, content_type, req.content_length,
                             [&](char* buf, size_t n) { return req.read_body(buf, n); });
   if (!ok) {
@@ -303,6 +328,10 @@ void Registry::blob_location(http::Request& req, http::ResponseWriter& w, const 
   std::map<std::string, std::string> properties;
   for (auto& kv : req.query) properties[kv.first] = kv.second;
   auto loc = store_->GetBlobLocation(name, digest, purpose, properties);
+  if (loc.supported) {
+    if (purpose == "upload") metrics_.presign_upload_total.fetch_add(1);
+    if (purpose == "download") metrics_.presign_download_total.fetch_add(1);
+  }
   if (!loc.supported) {
     response_error(w, wire::ErrorInfo{501, "UNSUPPORTED", "blob location not supported", ""});
     return;
@@ -317,6 +346,7 @@ void Registry::blob_location(http::Request& req, http::ResponseWriter& w, const 
 void Registry::garbage_collect(http::Request& req, http::ResponseWriter& w,
                                const std::string& name) {
   int removed = store_->GCBlobs(name);
+  metrics_.gc_blobs_removed_total.fetch_add(static_cast<uint64_t>(removed));
   json::Object o;
   o["blobs"] = json::Value(removed);
   response_ok(w, json::Value(std::move(o)));
