@@ -115,26 +115,103 @@ class GpuClient:
 
     # -------------------------------------------------------------- pull --
 
+    def _expected_leaves(self, repository: str, desc: types.Descriptor) -> Optional[bytes]:
+        """Fetch the leaves sidecar blob (per-chunk SHA-256 array) if the
+        descriptor carries one."""
+        ld = desc.annotations.get(types.ANNOTATION_LEAVES_BLOB, "")
+        if not ld:
+            return None
+        try:
+            data = b"".join(self.remote.get_blob_content(repository, ld))
+        except er.ModelxError:
+            return None
+        if dg.sha256_digest(data) != ld:
+            return None
+        return data
+
+    @staticmethod
+    def _bad_chunk_ranges(got: bytes, expect: bytes, chunk_size: int,
+                          size: int) -> List[Tuple[int, int]]:
+        """Contiguous (offset, length) ranges of mismatching chunks."""
+        n = len(expect) // 32
+        bad = [i for i in range(n) if got[i * 32 : (i + 1) * 32] != expect[i * 32 : (i + 1) * 32]]
+        ranges: List[Tuple[int, int]] = []
+        for i in bad:
+            off = i * chunk_size
+            ln = min(chunk_size, size - off)
+            if ranges and ranges[-1][0] + ranges[-1][1] == off:
+                ranges[-1] = (ranges[-1][0], ranges[-1][1] + ln)
+            else:
+                ranges.append((off, ln))
+        return ranges
+
+    def _fetch_ranges(self, url: str, headers: Dict[str, str], ptr: int,
+                      ranges: List[Tuple[int, int]]) -> int:
+        total = 0
+        for off, ln in ranges:
+            self.engine.pull_to_device(url, headers, ln, ptr + off, self.num_conns, off)
+            total += ln
+        return total
+
     def pull_blob_to_device(self, repository: str, desc: types.Descriptor,
-                            tensor=None, verify: bool = True) -> "torch.Tensor":
+                            tensor=None, verify: bool = True,
+                            resume: bool = False) -> "torch.Tensor":
+        """Land a blob in HBM. With ``resume=True`` and an existing tensor,
+        only chunks whose GPU hash mismatches the expected leaves are fetched
+        (chunk-level resume/dedup — the reference resumes at whole-blob
+        granularity only, pull.go:115-124). On digest mismatch after a full
+        pull, only the BAD chunks are refetched (SURVEY.md §5 fault-injection
+        requirement: a flipped byte must re-fetch the chunk, not the blob)."""
+        import time
+
         import torch
 
+        had_tensor = tensor is not None
         if tensor is None:
             tensor = torch.empty(desc.size, dtype=torch.uint8, device=f"cuda:{self.device}")
         assert tensor.numel() >= desc.size
-        import time
-
         url, headers = self._download_url(repository, desc)
+        cs = int(desc.annotations.get(types.ANNOTATION_CHUNK_SIZE, 0) or 0) or (
+            dg.algo_chunk_size(desc.digest.split(":", 1)[0]) or DEFAULT_GPU_CHUNK)
+
+        if resume and had_tensor:
+            expect = self._expected_leaves(repository, desc)
+            if expect is not None:
+                got = self.engine.sha256_chunk_leaves(tensor.data_ptr(), desc.size, cs)
+                ranges = self._bad_chunk_ranges(got, expect, cs, desc.size)
+                fetched = self._fetch_ranges(url, headers, tensor.data_ptr(), ranges)
+                self.last_stats.append({"phase": "pull-resume", "bytes": fetched,
+                                        "skipped": desc.size - fetched})
+                if verify:
+                    self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+                return tensor
+            # no leaves sidecar → fall through to a full pull
+
         stats = self.engine.pull_to_device(url, headers, desc.size, tensor.data_ptr(),
                                            self.num_conns)
         stats["name"] = desc.name
         stats["phase"] = "pull-transfer"
         self.last_stats.append(stats)
-        if verify:
-            t0 = time.monotonic()
+        if not verify:
+            return tensor
+        t0 = time.monotonic()
+        try:
             self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
-            self.last_stats.append({"phase": "pull-verify", "bytes": desc.size,
-                                    "seconds": time.monotonic() - t0})
+        except er.ModelxError:
+            # chunk-level refetch before giving up
+            expect = self._expected_leaves(repository, desc)
+            if expect is None:
+                raise
+            got = self.engine.sha256_chunk_leaves(tensor.data_ptr(), desc.size, cs)
+            ranges = self._bad_chunk_ranges(got, expect, cs, desc.size)
+            if not ranges:
+                raise
+            self._fetch_ranges(url, headers, tensor.data_ptr(), ranges)
+            self.last_stats.append({"phase": "pull-chunk-refetch",
+                                    "bytes": sum(r[1] for r in ranges)})
+            self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+        self.last_stats.append({"phase": "pull-verify", "bytes": desc.size,
+                                "seconds": time.monotonic() - t0})
         return tensor
 
     def pull_to_gpu(self, repository: str, version: str = "",
@@ -144,7 +221,7 @@ class GpuClient:
         manifest = self.remote.get_manifest(repository, version)
         out: Dict[str, "torch.Tensor"] = {}
         for desc in manifest.blobs:
-            if desc.size == 0:
+            if desc.size == 0 or desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
                 continue
             out[desc.name] = self.pull_blob_to_device(repository, desc, verify=verify)
         return out
@@ -154,6 +231,12 @@ class GpuClient:
     def digest_device_blob(self, ptr: int, size: int,
                            chunk_size: int = DEFAULT_GPU_CHUNK) -> Tuple[str, str]:
         """(chunked_digest, chunk_digest_annotation) of device memory."""
+        root, _ = self.digest_device_blob_with_leaves(ptr, size, chunk_size)
+        return root, root
+
+    def digest_device_blob_with_leaves(self, ptr: int, size: int,
+                                       chunk_size: int = DEFAULT_GPU_CHUNK
+                                       ) -> Tuple[str, bytes]:
         import time
 
         t0 = time.monotonic()
@@ -161,7 +244,18 @@ class GpuClient:
         root = dg.root_from_leaf_bytes(leaves, chunk_size, size)
         self.last_stats.append({"phase": "gpu-digest", "bytes": size,
                                 "seconds": time.monotonic() - t0})
-        return root, root
+        return root, leaves
+
+    def _upload_small(self, repository: str, desc: types.Descriptor, data: bytes) -> None:
+        loc = self.remote.get_blob_location(repository, desc, "upload")
+        if loc is not None:
+            import requests
+
+            p = (loc.properties.get("parts") or [{}])[0]
+            requests.request(p.get("method") or "PUT", p["url"],
+                             headers=_signed_headers(p), data=data).raise_for_status()
+        else:
+            self.remote.upload_blob_content(repository, desc, data)
 
     def push_blob_from_device(self, repository: str, desc: types.Descriptor, ptr: int,
                               part_bytes: int = DEFAULT_PART_BYTES,
@@ -220,25 +314,28 @@ class GpuClient:
             name="modelx.yaml", media_type=types.MEDIA_TYPE_MODEL_CONFIG_YAML,
             digest=cfg_digest, size=len(cfg), modified=now)
         if not self.remote.head_blob(repository, cfg_digest):
-            loc = self.remote.get_blob_location(repository, manifest.config, "upload")
-            if loc is not None:
-                import requests
-
-                p = (loc.properties.get("parts") or [{}])[0]
-                requests.request(p.get("method") or "PUT", p["url"],
-                                 headers=_signed_headers(p), data=cfg).raise_for_status()
-            else:
-                self.remote.upload_blob_content(repository, manifest.config, cfg)
+            self._upload_small(repository, manifest.config, cfg)
         for name, t in tensors.items():
             size = t.numel() * t.element_size()
-            root, note = self.digest_device_blob(t.data_ptr(), size, chunk_size)
+            root, leaves = self.digest_device_blob_with_leaves(t.data_ptr(), size, chunk_size)
+            leaves_digest = dg.sha256_digest(leaves)
             desc = types.Descriptor(
                 name=name, media_type=types.MEDIA_TYPE_MODEL_FILE, digest=root, size=size,
                 modified=now,
-                annotations={types.ANNOTATION_CHUNK_DIGEST: note,
-                             types.ANNOTATION_CHUNK_SIZE: str(chunk_size)})
+                annotations={types.ANNOTATION_CHUNK_DIGEST: root,
+                             types.ANNOTATION_CHUNK_SIZE: str(chunk_size),
+                             types.ANNOTATION_LEAVES_BLOB: leaves_digest})
             self.push_blob_from_device(repository, desc, t.data_ptr(), part_bytes=part_bytes)
             manifest.blobs.append(desc)
+            # leaves sidecar: 32 B per chunk, enables chunk-level
+            # resume/refetch/dedup on pull; listed in the manifest so GC
+            # keeps it, with its own media type so clients can skip it
+            ldesc = types.Descriptor(
+                name=name + ".leaves", media_type=types.MEDIA_TYPE_MODEL_LEAVES,
+                digest=leaves_digest, size=len(leaves), modified=now)
+            if not self.remote.head_blob(repository, leaves_digest):
+                self._upload_small(repository, ldesc, leaves)
+            manifest.blobs.append(ldesc)
         manifest.blobs = types.sort_descriptors_by_name(manifest.blobs)
         self.remote.put_manifest(repository, version or "latest", manifest)
         return manifest

@@ -37,6 +37,11 @@ MEDIA_TYPE_MODEL_DIRECTORY_TARGZ = "application/vnd.modelx.model.directory.v1.ta
 ANNOTATION_CHUNK_DIGEST = "modelx.amd/chunk-digest"
 #   chunk size used for the chunked digest, bytes (decimal string)
 ANNOTATION_CHUNK_SIZE = "modelx.amd/chunk-size"
+#   digest of the companion leaves blob (packed 32 B per-chunk SHA-256 array)
+#   enabling chunk-level resume/refetch/dedup
+ANNOTATION_LEAVES_BLOB = "modelx.amd/leaves-blob"
+# media type of a leaves blob (listed in the manifest so GC keeps it)
+MEDIA_TYPE_MODEL_LEAVES = "application/vnd.modelx.amd.leaves.v1"
 
 GO_ZERO_TIME = "0001-01-01T00:00:00Z"
 

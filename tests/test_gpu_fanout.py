@@ -77,3 +77,42 @@ def test_fanout_sharded_world1(stack):
             assert torch.equal(out[name], t)
     finally:
         dist.destroy_process_group()
+
+
+def test_chunk_level_resume_fetches_only_bad_chunks(stack):
+    """Corrupt 2 MiB of a 32 MiB resident copy → resume fetches only the bad
+    chunks, not the blob (SURVEY.md §5 fault-injection requirement)."""
+    from modelx_amd.client.gpu import GpuClient
+
+    mdx, _ = stack
+    g = GpuClient(mdx.url, device=0)
+    src = torch.randint(0, 256, (32 << 20,), dtype=torch.uint8, device="cuda:0")
+    g.push_from_gpu("fan/resume", "v1", {"w.bin": src})
+    desc = next(b for b in g.remote.get_manifest("fan/resume", "v1").blobs
+                if b.name == "w.bin")
+    # local copy with a 2 MiB hole
+    local = src.clone()
+    local[5 << 20 : 7 << 20] = 0
+    g.last_stats.clear()
+    out = g.pull_blob_to_device("fan/resume", desc, tensor=local, resume=True)
+    assert torch.equal(out, src)
+    st = next(s for s in g.last_stats if s.get("phase") == "pull-resume")
+    assert st["bytes"] <= 3 << 20  # only the corrupted region (+ boundary chunks)
+    assert st["skipped"] >= 29 << 20
+
+
+def test_leaves_sidecar_in_manifest(stack):
+    from modelx_amd.client.gpu import GpuClient
+    from modelx_amd.wire import types as wt
+
+    mdx, _ = stack
+    g = GpuClient(mdx.url, device=0)
+    src = torch.randint(0, 256, (4 << 20,), dtype=torch.uint8, device="cuda:0")
+    m = g.push_from_gpu("fan/leaves", "v1", {"w.bin": src})
+    names = {b.name: b for b in m.blobs}
+    assert "w.bin.leaves" in names
+    assert names["w.bin.leaves"].media_type == wt.MEDIA_TYPE_MODEL_LEAVES
+    assert names["w.bin"].annotations[wt.ANNOTATION_LEAVES_BLOB] == names["w.bin.leaves"].digest
+    # pull_to_gpu must skip the sidecar
+    out = g.pull_to_gpu("fan/leaves", "v1")
+    assert set(out) == {"w.bin"}
