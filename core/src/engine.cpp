@@ -46,6 +46,19 @@ extern "C" hipError_t modelx_sha256_chunk_leaves(const void* data, uint64_t tota
                                                  uint32_t nchunks, hipStream_t stream);
 extern "C" hipError_t modelx_sha256_multibuf(const void* const* buffers, const uint64_t* lengths,
                                              uint32_t nbuf, void* digests, hipStream_t stream);
+extern "C" hipError_t modelx_tar_index(const void* tar, uint64_t tar_len, void* entries,
+                                       uint32_t max_entries, uint32_t* count_dev,
+                                       uint32_t* error_dev, hipStream_t stream);
+extern "C" hipError_t modelx_tar_scatter(const void* tar, const void* segs, uint32_t nsegs,
+                                         hipStream_t stream);
+
+struct TarEntryHost {
+  uint64_t header_off, payload_off, size;
+  uint32_t typeflag, mode;
+};
+struct CopySegHost {
+  uint64_t src_off, dst_ptr, len;
+};
 
 #define HIP_CHECK(expr)                                                                 \
   do {                                                                                  \
@@ -343,6 +356,122 @@ class GpuEngine {
     return stats;
   }
 
+  // ---------------------------------------------------- tar (directories) --
+
+  // Index a tar archive resident in HBM. Returns a list of
+  // (name, payload_off, size, mode) for regular files (GNU longnames
+  // resolved). Runs the sequential header walk on-device, gathers the header
+  // blocks with the scatter kernel, and reads names from one D2H copy.
+  py::list tar_index(uintptr_t tar_ptr, uint64_t tar_len) {
+    HIP_CHECK(hipSetDevice(device_));
+    uint32_t max_entries = 65536;
+    void* dentries = nullptr;
+    uint32_t* dcount = nullptr;
+    HIP_CHECK(hipMalloc(&dentries, max_entries * sizeof(TarEntryHost)));
+    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dcount), 2 * sizeof(uint32_t)));
+    std::vector<TarEntryHost> entries;
+    uint32_t count = 0, error = 0;
+    {
+      py::gil_scoped_release release;
+      HIP_CHECK(modelx_tar_index(reinterpret_cast<void*>(tar_ptr), tar_len, dentries,
+                                 max_entries, dcount, dcount + 1, hash_stream_));
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+      HIP_CHECK(hipMemcpy(&count, dcount, sizeof count, hipMemcpyDeviceToHost));
+      HIP_CHECK(hipMemcpy(&error, dcount + 1, sizeof error, hipMemcpyDeviceToHost));
+      if (error) {
+        hipFree(dentries);
+        hipFree(dcount);
+        throw std::runtime_error("tar_index: malformed archive or too many entries");
+      }
+      entries.resize(count);
+      if (count)
+        HIP_CHECK(hipMemcpy(entries.data(), dentries, count * sizeof(TarEntryHost),
+                            hipMemcpyDeviceToHost));
+      hipFree(dentries);
+      hipFree(dcount);
+    }
+    // gather headers (+ longname payloads) to host in one shot
+    std::string headers(static_cast<size_t>(count) * 512, '\0');
+    if (count) {
+      py::gil_scoped_release release;
+      void* dgather = nullptr;
+      HIP_CHECK(hipMalloc(&dgather, headers.size()));
+      std::vector<CopySegHost> segs(count);
+      for (uint32_t i = 0; i < count; i++)
+        segs[i] = {entries[i].header_off, reinterpret_cast<uint64_t>(dgather) + i * 512, 512};
+      void* dsegs = nullptr;
+      HIP_CHECK(hipMalloc(&dsegs, segs.size() * sizeof(CopySegHost)));
+      HIP_CHECK(hipMemcpy(dsegs, segs.data(), segs.size() * sizeof(CopySegHost),
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(modelx_tar_scatter(reinterpret_cast<void*>(tar_ptr), dsegs, count,
+                                   hash_stream_));
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+      HIP_CHECK(hipMemcpy(&headers[0], dgather, headers.size(), hipMemcpyDeviceToHost));
+      hipFree(dsegs);
+      hipFree(dgather);
+    }
+    py::list out;
+    std::string pending_longname;
+    for (uint32_t i = 0; i < count; i++) {
+      const char* hdr = headers.data() + static_cast<size_t>(i) * 512;
+      uint32_t tf = entries[i].typeflag;
+      if (tf == 'L') {
+        // GNU longname: payload holds the next entry's name
+        std::string ln(static_cast<size_t>(entries[i].size), '\0');
+        if (!ln.empty()) {
+          py::gil_scoped_release release;
+          HIP_CHECK(hipMemcpy(&ln[0], reinterpret_cast<char*>(tar_ptr) + entries[i].payload_off,
+                              ln.size(), hipMemcpyDeviceToHost));
+        }
+        while (!ln.empty() && ln.back() == '\0') ln.pop_back();
+        pending_longname = ln;
+        continue;
+      }
+      std::string name;
+      if (!pending_longname.empty()) {
+        name = pending_longname;
+        pending_longname.clear();
+      } else {
+        name.assign(hdr, strnlen(hdr, 100));
+        // ustar prefix field (bytes 345..500)
+        std::string prefix(hdr + 345, strnlen(hdr + 345, 155));
+        if (!prefix.empty()) name = prefix + "/" + name;
+      }
+      if (tf != '0' && tf != 0) continue;  // regular files only
+      py::dict e;
+      e["name"] = name;
+      e["offset"] = entries[i].payload_off;
+      e["size"] = entries[i].size;
+      e["mode"] = entries[i].mode;
+      out.append(std::move(e));
+    }
+    return out;
+  }
+
+  // Scatter payload segments: [(src_off, dst_dev_ptr, len), ...]
+  void tar_scatter(uintptr_t tar_ptr,
+                   const std::vector<std::tuple<uint64_t, uintptr_t, uint64_t>>& segs) {
+    HIP_CHECK(hipSetDevice(device_));
+    if (segs.empty()) return;
+    py::gil_scoped_release release;
+    // split into <=4 MiB pieces for load balance across CUs
+    std::vector<CopySegHost> pieces;
+    constexpr uint64_t kPiece = 4ull << 20;
+    for (auto& t : segs) {
+      uint64_t src_off = std::get<0>(t), dst = std::get<1>(t), len = std::get<2>(t);
+      for (uint64_t off = 0; off < len; off += kPiece)
+        pieces.push_back({src_off + off, dst + off, std::min(kPiece, len - off)});
+    }
+    void* dsegs = nullptr;
+    HIP_CHECK(hipMalloc(&dsegs, pieces.size() * sizeof(CopySegHost)));
+    HIP_CHECK(hipMemcpy(dsegs, pieces.data(), pieces.size() * sizeof(CopySegHost),
+                        hipMemcpyHostToDevice));
+    HIP_CHECK(modelx_tar_scatter(reinterpret_cast<void*>(tar_ptr), dsegs,
+                                 static_cast<uint32_t>(pieces.size()), hash_stream_));
+    HIP_CHECK(hipStreamSynchronize(hash_stream_));
+    hipFree(dsegs);
+  }
+
   void synchronize() {
     HIP_CHECK(hipSetDevice(device_));
     for (auto& st : streams_) HIP_CHECK(hipStreamSynchronize(st));
@@ -426,5 +555,7 @@ PYBIND11_MODULE(_core, m) {
       .def("sha256_multibuf", &GpuEngine::sha256_multibuf, py::arg("buffers"))
       .def("push_part_from_device", &GpuEngine::push_part_from_device, py::arg("url"),
            py::arg("method"), py::arg("headers"), py::arg("src_ptr"), py::arg("size"))
+      .def("tar_index", &GpuEngine::tar_index, py::arg("tar_ptr"), py::arg("tar_len"))
+      .def("tar_scatter", &GpuEngine::tar_scatter, py::arg("tar_ptr"), py::arg("segs"))
       .def("synchronize", &GpuEngine::synchronize);
 }

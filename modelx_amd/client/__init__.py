@@ -23,9 +23,9 @@ class Client:
 
     def push(self, repository: str, version: str, basedir: str,
              configfile: str = "modelx.yaml", digest_mode: str = "sha256",
-             quiet: Optional[bool] = None) -> types.Manifest:
+             quiet: Optional[bool] = None, dir_format: str = "tar+gz") -> types.Manifest:
         return self.pusher.push(repository, version or "latest", basedir, configfile,
-                                digest_mode=digest_mode, quiet=quiet)
+                                digest_mode=digest_mode, quiet=quiet, dir_format=dir_format)
 
     def pull(self, repository: str, version: str, into_dir: str,
              quiet: Optional[bool] = None) -> types.Manifest:

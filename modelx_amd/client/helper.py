@@ -26,18 +26,20 @@ class _TeeWriter(io.RawIOBase):
         return len(b)
 
 
-def tgz(src_dir: str, out_file: Optional[str], chunk_size: int = 1 << 20
-        ) -> Tuple[str, str, int]:
-    """Archive src_dir to tar.gz (optionally writing out_file), digesting the
-    compressed stream. Returns (sha256_digest, chunk_digest, size).
-    Deterministic: entries sorted, mtime preserved (like the reference's tar)."""
+def tgz(src_dir: str, out_file: Optional[str], chunk_size: int = 1 << 20,
+        compress: bool = True) -> Tuple[str, str, int]:
+    """Archive src_dir to tar[.gz] (optionally writing out_file), digesting
+    the output stream. Returns (sha256_digest, chunk_digest, size).
+    Deterministic: entries sorted, gzip mtime zeroed → digests stable.
+    compress=False emits plain tar (GPU-scatterable, see core/hip/tar.hip)."""
     digester = StreamingDigester(chunk_size=chunk_size)
     sink = open(out_file, "wb") if out_file else None
     try:
         tee = _TeeWriter(sink, digester)
-        # gzip with mtime=0 so archives are reproducible → digests stable
-        with gzip.GzipFile(fileobj=tee, mode="wb", mtime=0) as gz:
-            with tarfile.open(fileobj=gz, mode="w|") as tar:
+        ctx = gzip.GzipFile(fileobj=tee, mode="wb", mtime=0) if compress else None
+        stream = ctx if compress else tee
+        try:
+            with tarfile.open(fileobj=stream, mode="w|") as tar:
                 base = os.path.basename(src_dir.rstrip("/"))
                 entries = []
                 for root, dirs, files in os.walk(src_dir):
@@ -47,17 +49,24 @@ def tgz(src_dir: str, out_file: Optional[str], chunk_size: int = 1 << 20
                 for path in entries:
                     arcname = os.path.join(base, os.path.relpath(path, src_dir))
                     tar.add(path, arcname=arcname, recursive=False)
+        finally:
+            if ctx is not None:
+                ctx.close()
     finally:
         if sink:
             sink.close()
     return digester.canonical_digest(), digester.chunk_digest(), digester.total
 
 
-def untgz(archive_path: str, dest_dir: str) -> None:
-    """Extract tar.gz stripping the top-level directory component
+def tar_plain(src_dir: str, out_file: Optional[str], chunk_size: int = 1 << 20):
+    return tgz(src_dir, out_file, chunk_size, compress=False)
+
+
+def untgz(archive_path: str, dest_dir: str, compressed: bool = True) -> None:
+    """Extract tar[.gz] stripping the top-level directory component
     (reference: helper.go:55-79 extracts into dest)."""
     os.makedirs(dest_dir, exist_ok=True)
-    with tarfile.open(archive_path, mode="r:gz") as tar:
+    with tarfile.open(archive_path, mode="r:gz" if compressed else "r:") as tar:
         for member in tar.getmembers():
             parts = member.name.split("/", 1)
             member.name = parts[1] if len(parts) == 2 else parts[0]

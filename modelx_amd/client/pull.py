@@ -56,8 +56,12 @@ class Puller:
 
     def _pull_one(self, repository: str, desc: types.Descriptor, into_dir: str,
                   bar: Optional[Bar]) -> None:
-        if desc.media_type == types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ:
+        if desc.media_type in (types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ,
+                               types.MEDIA_TYPE_MODEL_DIRECTORY_TAR):
             self._pull_directory(repository, desc, into_dir, bar)
+        elif desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
+            if bar:
+                bar.set_status("sidecar", complete=True)
         else:
             self._pull_file(repository, desc, into_dir, bar)
 
@@ -79,23 +83,25 @@ class Puller:
 
     def _pull_directory(self, repository: str, desc: types.Descriptor, into_dir: str,
                         bar: Optional[Bar]) -> None:
-        from .helper import digest_tgz_of_dir, untgz
+        from .helper import tgz, untgz
 
+        compressed = desc.media_type == types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ
         target = os.path.join(into_dir, desc.name)
         if os.path.isdir(target):
             # re-archive locally and compare digests (pull.go:148-154)
-            canonical, chunked, _ = digest_tgz_of_dir(target)
+            canonical, chunked, _ = tgz(target, None, compress=compressed)
             if desc.digest in (canonical, chunked):
                 if bar:
                     bar.set_status("up to date", complete=True)
                 return
         cache_dir = os.path.join(into_dir, MODELX_CACHE_DIR)
         os.makedirs(cache_dir, exist_ok=True)
-        tgz_path = os.path.join(cache_dir, desc.name + ".tar.gz")
+        suffix = ".tar.gz" if compressed else ".tar"
+        tgz_path = os.path.join(cache_dir, desc.name + suffix)
         # cached two-phase: download then extract (pull.go:158-182)
         if not (os.path.isfile(tgz_path) and _verify_digest_of_file(tgz_path, desc.digest)):
             self.pull_blob(repository, desc, tgz_path, bar)
-        untgz(tgz_path, target)
+        untgz(tgz_path, target, compressed=compressed)
 
     # -------------------------------------------------------------- blobs --
 

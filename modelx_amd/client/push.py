@@ -53,7 +53,8 @@ def _file_descriptor(path: str, name: str, chunk_size: int, digest_mode: str) ->
 
 def parse_manifest(basedir: str, configfile: str = "modelx.yaml",
                    chunk_size: int = dg.DEFAULT_CHUNK_SIZE,
-                   digest_mode: str = "sha256") -> types.Manifest:
+                   digest_mode: str = "sha256",
+                   dir_format: str = "tar+gz") -> types.Manifest:
     """Scan basedir into a Manifest (reference: push.go:67-100).
     Dot-files skipped; directories become tar.gz descriptors."""
     manifest = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
@@ -70,14 +71,17 @@ def parse_manifest(basedir: str, configfile: str = "modelx.yaml",
             manifest.config = desc
             continue
         if os.path.isdir(path):
-            # dirs → tar.gz blob (push.go:86-92,102-118), archived to cache
+            # dirs → tar[.gz] blob (push.go:86-92,102-118), archived to cache
             from .helper import tgz
 
-            tgz_path = os.path.join(cache_dir, entry + ".tar.gz")
-            canonical, chunked, size = tgz(path, tgz_path, chunk_size)
+            compressed = dir_format != "tar"
+            suffix = ".tar.gz" if compressed else ".tar"
+            tgz_path = os.path.join(cache_dir, entry + suffix)
+            canonical, chunked, size = tgz(path, tgz_path, chunk_size, compress=compressed)
             desc = types.Descriptor(
                 name=entry,
-                media_type=types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ,
+                media_type=types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ if compressed
+                else types.MEDIA_TYPE_MODEL_DIRECTORY_TAR,
                 digest=chunked if digest_mode == "chunked" else canonical,
                 size=size,
                 modified=datetime.fromtimestamp(os.stat(path).st_mtime, tz=timezone.utc),
@@ -102,9 +106,9 @@ class Pusher:
 
     def push(self, repository: str, version: str, basedir: str,
              configfile: str = "modelx.yaml", digest_mode: str = "sha256",
-             chunk_size: int = dg.DEFAULT_CHUNK_SIZE, quiet: Optional[bool] = None
-             ) -> types.Manifest:
-        manifest = parse_manifest(basedir, configfile, chunk_size, digest_mode)
+             chunk_size: int = dg.DEFAULT_CHUNK_SIZE, quiet: Optional[bool] = None,
+             dir_format: str = "tar+gz") -> types.Manifest:
+        manifest = parse_manifest(basedir, configfile, chunk_size, digest_mode, dir_format)
         with MultiBar(f"push {repository}@{version}", self.concurrency, quiet=quiet) as mb:
             descs: List[types.Descriptor] = [manifest.config] + list(manifest.blobs)
             for desc in descs:
@@ -120,6 +124,8 @@ class Pusher:
     def _blob_source(basedir: str, desc: types.Descriptor) -> str:
         if desc.media_type == types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ:
             return os.path.join(basedir, MODELX_CACHE_DIR, desc.name + ".tar.gz")
+        if desc.media_type == types.MEDIA_TYPE_MODEL_DIRECTORY_TAR:
+            return os.path.join(basedir, MODELX_CACHE_DIR, desc.name + ".tar")
         return os.path.join(basedir, desc.name)
 
     def push_blob(self, repository: str, desc: types.Descriptor, src_path: str,

@@ -42,6 +42,10 @@ ANNOTATION_CHUNK_SIZE = "modelx.amd/chunk-size"
 ANNOTATION_LEAVES_BLOB = "modelx.amd/leaves-blob"
 # media type of a leaves blob (listed in the manifest so GC keeps it)
 MEDIA_TYPE_MODEL_LEAVES = "application/vnd.modelx.amd.leaves.v1"
+# uncompressed tar directory blob — GPU-scatterable without a CPU inflate
+# (optional alternative to ...directory.v1.tar+gz; reference clients treat it
+# as an opaque file)
+MEDIA_TYPE_MODEL_DIRECTORY_TAR = "application/vnd.modelx.amd.directory.v1.tar"
 
 GO_ZERO_TIME = "0001-01-01T00:00:00Z"
 

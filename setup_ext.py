@@ -17,6 +17,7 @@ ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 SOURCES = [
     "core/src/engine.cpp",
     "core/hip/sha256.hip",
+    "core/hip/tar.hip",
     "core/src/http.cpp",
     "core/src/json.cpp",
 ]
