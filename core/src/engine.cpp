@@ -412,21 +412,42 @@ class GpuEngine {
     }
     py::list out;
     std::string pending_longname;
+    auto read_payload = [&](uint32_t i) {
+      std::string pl(static_cast<size_t>(entries[i].size), '\0');
+      if (!pl.empty()) {
+        py::gil_scoped_release release;
+        HIP_CHECK(hipMemcpy(&pl[0], reinterpret_cast<char*>(tar_ptr) + entries[i].payload_off,
+                            pl.size(), hipMemcpyDeviceToHost));
+      }
+      return pl;
+    };
     for (uint32_t i = 0; i < count; i++) {
       const char* hdr = headers.data() + static_cast<size_t>(i) * 512;
       uint32_t tf = entries[i].typeflag;
       if (tf == 'L') {
         // GNU longname: payload holds the next entry's name
-        std::string ln(static_cast<size_t>(entries[i].size), '\0');
-        if (!ln.empty()) {
-          py::gil_scoped_release release;
-          HIP_CHECK(hipMemcpy(&ln[0], reinterpret_cast<char*>(tar_ptr) + entries[i].payload_off,
-                              ln.size(), hipMemcpyDeviceToHost));
-        }
+        std::string ln = read_payload(i);
         while (!ln.empty() && ln.back() == '\0') ln.pop_back();
         pending_longname = ln;
         continue;
       }
+      if (tf == 'x' || tf == 'X') {
+        // PAX extended header: "<len> key=value\n" records; path overrides
+        // the next entry's name (what Python tarfile emits for long names)
+        std::string px = read_payload(i);
+        size_t pos = 0;
+        while (pos < px.size()) {
+          size_t sp = px.find(' ', pos);
+          if (sp == std::string::npos) break;
+          long rec_len = atol(px.c_str() + pos);
+          if (rec_len <= 0 || pos + static_cast<size_t>(rec_len) > px.size()) break;
+          std::string rec = px.substr(sp + 1, pos + rec_len - sp - 2);  // drop trailing \n
+          if (rec.rfind("path=", 0) == 0) pending_longname = rec.substr(5);
+          pos += static_cast<size_t>(rec_len);
+        }
+        continue;
+      }
+      if (tf == 'g') continue;  // pax global header
       std::string name;
       if (!pending_longname.empty()) {
         name = pending_longname;
