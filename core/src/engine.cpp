@@ -120,6 +120,35 @@ class GpuEngine {
   py::dict pull_to_device(const std::string& url,
                           const std::map<std::string, std::string>& headers, uint64_t size,
                           uintptr_t dst_ptr, int num_conns, uint64_t base_offset = 0) {
+    std::string leaves;
+    return pull_impl(url, headers, size, dst_ptr, num_conns, base_offset, 0, &leaves);
+  }
+
+  // Same, but ALSO hashes every landed slot with the CDNA4 SHA-256 chunk
+  // kernel on the slot's own stream right after its H2D copy — verification
+  // overlaps the remaining transfer ("verify as chunks land"). Returns
+  // (stats, leaves). slot_bytes must be a chunk_size multiple (64 MiB / 128
+  // KiB is), so every slot covers whole chunks.
+  py::tuple pull_to_device_hashed(const std::string& url,
+                                  const std::map<std::string, std::string>& headers,
+                                  uint64_t size, uintptr_t dst_ptr, int num_conns,
+                                  uint64_t chunk_size) {
+    std::string leaves;
+    py::dict stats = pull_impl(url, headers, size, dst_ptr, num_conns, 0, chunk_size, &leaves);
+    return py::make_tuple(stats, py::bytes(leaves));
+  }
+
+  py::dict pull_impl(const std::string& url, const std::map<std::string, std::string>& headers,
+                     uint64_t size, uintptr_t dst_ptr, int num_conns, uint64_t base_offset,
+                     uint64_t hash_chunk, std::string* leaves_out) {
+    void* dleaves = nullptr;
+    uint32_t total_chunks = 0;
+    if (hash_chunk) {
+      if (slot_bytes_ % hash_chunk != 0)
+        throw std::runtime_error("slot_bytes must be a multiple of chunk_size");
+      total_chunks = size ? static_cast<uint32_t>((size + hash_chunk - 1) / hash_chunk) : 1;
+      HIP_CHECK(hipMalloc(&dleaves, static_cast<size_t>(total_chunks) * 32));
+    }
     py::gil_scoped_release release;
     HIP_CHECK(hipSetDevice(device_));
     double t0 = now_s();
@@ -197,6 +226,16 @@ class GpuEngine {
           hipStream_t st = streams_[stream_rr.fetch_add(1) % streams_.size()];
           hipError_t e = hipMemcpyAsync(reinterpret_cast<char*>(dst_ptr) + r.offset, slot->host,
                                         r.length, hipMemcpyHostToDevice, st);
+          if (e == hipSuccess && hash_chunk) {
+            // hash this slot's chunks right behind its copy (stream-ordered)
+            uint32_t first_chunk = static_cast<uint32_t>(r.offset / hash_chunk);
+            uint32_t n_chunks = static_cast<uint32_t>((r.length + hash_chunk - 1) / hash_chunk);
+            uint64_t span = std::min<uint64_t>(r.length, size - r.offset);
+            e = modelx_sha256_chunk_leaves(
+                reinterpret_cast<char*>(dst_ptr) + r.offset, span, hash_chunk,
+                static_cast<char*>(dleaves) + static_cast<size_t>(first_chunk) * 32, n_chunks,
+                st);
+          }
           if (e == hipSuccess) e = hipEventRecord(slot->event, st);
           if (e != hipSuccess) {
             std::lock_guard<std::mutex> lk(mu_);
@@ -217,6 +256,12 @@ class GpuEngine {
     cv_pending_.notify_all();
     reclaimer.join();
     for (auto& st : streams_) HIP_CHECK(hipStreamSynchronize(st));
+    if (hash_chunk) {
+      leaves_out->resize(static_cast<size_t>(total_chunks) * 32);
+      HIP_CHECK(hipMemcpy(&(*leaves_out)[0], dleaves, leaves_out->size(),
+                          hipMemcpyDeviceToHost));
+      hipFree(dleaves);
+    }
     double t1 = now_s();
     {
       std::lock_guard<std::mutex> lk(mu_);
@@ -571,6 +616,9 @@ PYBIND11_MODULE(_core, m) {
       .def("pull_to_device", &GpuEngine::pull_to_device, py::arg("url"), py::arg("headers"),
            py::arg("size"), py::arg("dst_ptr"), py::arg("num_conns") = 8,
            py::arg("base_offset") = 0)
+      .def("pull_to_device_hashed", &GpuEngine::pull_to_device_hashed, py::arg("url"),
+           py::arg("headers"), py::arg("size"), py::arg("dst_ptr"), py::arg("num_conns") = 8,
+           py::arg("chunk_size") = (uint64_t)(128 << 10))
       .def("sha256_chunk_leaves", &GpuEngine::sha256_chunk_leaves, py::arg("dev_ptr"),
            py::arg("size"), py::arg("chunk_size"))
       .def("sha256_multibuf", &GpuEngine::sha256_multibuf, py::arg("buffers"))

@@ -69,6 +69,7 @@ class GpuClient:
         self.remote = RegistryClient(registry, authorization)
         self.device = device
         self.num_conns = num_conns
+        self.slot_bytes = slot_bytes
         self.engine = core.GpuEngine(device=device, num_slots=num_slots,
                                      slot_bytes=slot_bytes, num_streams=4)
         self.last_stats: List[dict] = []
@@ -187,11 +188,35 @@ class GpuClient:
                 return tensor
             # no leaves sidecar → fall through to a full pull
 
-        stats = self.engine.pull_to_device(url, headers, desc.size, tensor.data_ptr(),
-                                           self.num_conns)
-        stats["name"] = desc.name
-        stats["phase"] = "pull-transfer"
-        self.last_stats.append(stats)
+        # streaming-hash path: chunks are hashed on each slot's stream right
+        # behind its H2D copy, so verification overlaps the transfer
+        expect_digest = ""
+        algo_cs = dg.algo_chunk_size(desc.digest.split(":", 1)[0]) if desc.digest else None
+        if algo_cs:
+            expect_digest = desc.digest
+        elif desc.annotations.get(types.ANNOTATION_CHUNK_DIGEST, ""):
+            expect_digest = desc.annotations[types.ANNOTATION_CHUNK_DIGEST]
+        stream_hash = (verify and expect_digest
+                       and self.slot_bytes % cs == 0 and desc.size > 0)
+        t0 = time.monotonic()
+        if stream_hash:
+            stats, leaves = self.engine.pull_to_device_hashed(
+                url, headers, desc.size, tensor.data_ptr(), self.num_conns, cs)
+            stats["name"] = desc.name
+            stats["phase"] = "pull-transfer-hashed"
+            self.last_stats.append(stats)
+            got = dg.root_from_leaf_bytes(leaves, cs, desc.size)
+            if got == expect_digest:
+                self.last_stats.append({"phase": "pull-verify", "bytes": desc.size,
+                                        "seconds": time.monotonic() - t0 - stats["seconds"]})
+                return tensor
+            # fall through to the refetch path below
+        else:
+            stats = self.engine.pull_to_device(url, headers, desc.size, tensor.data_ptr(),
+                                               self.num_conns)
+            stats["name"] = desc.name
+            stats["phase"] = "pull-transfer"
+            self.last_stats.append(stats)
         if not verify:
             return tensor
         t0 = time.monotonic()
