@@ -111,29 +111,34 @@ def main():
     device = local_rank
     torch.cuda.set_device(device)
 
-    # --- shared local S3 + registry stack (rank 0 owns the processes) ------
+    # --- per-rank S3 + registry stack -------------------------------------
+    # Each rank runs its own modelxd + s3d shard (object stores scale
+    # horizontally; the measured entity is the per-GPU client data plane).
     from util_servers import MODELXD, S3D, ServerProc, wait_http
 
     master_port = int(os.environ.get("MASTER_PORT", "29500"))
-    s3_port = master_port + 1371
-    mdx_port = master_port + 1372
-    store_root = pick_store_root(args)
+    s3_port = master_port + 1371 + 2 * rank
+    mdx_port = master_port + 1372 + 2 * rank
+    store_root = pick_store_root(args) + f"-r{rank}"
     procs = []
-    if rank == 0:
-        shutil.rmtree(store_root, ignore_errors=True)
-        os.makedirs(os.path.join(store_root, "modelx"), exist_ok=True)
-        if not (os.path.exists(S3D) and os.path.exists(MODELXD)):
-            import subprocess
+    shutil.rmtree(store_root, ignore_errors=True)
+    os.makedirs(os.path.join(store_root, "modelx"), exist_ok=True)
+    if rank == 0 and not (os.path.exists(S3D) and os.path.exists(MODELXD)):
+        import subprocess
 
-            subprocess.run(["make", "servers"], cwd=REPO, check=True)
-        procs.append(ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root", store_root,
-                                 "--access-key", "modelx", "--secret-key", "modelx123"], s3_port))
-        wait_http(s3_port)
-        procs.append(ServerProc(
-            [MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
-             f"http://127.0.0.1:{s3_port}", "--s3-bucket", "modelx", "--s3-access-key",
-             "modelx", "--s3-secret-key", "modelx123", "--enable-redirect"], mdx_port))
-        wait_http(mdx_port)
+        subprocess.run(["make", "servers"], cwd=REPO, check=True)
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()  # wait for a possible rank-0 build
+    procs.append(ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root", store_root,
+                             "--access-key", "modelx", "--secret-key", "modelx123"], s3_port))
+    wait_http(s3_port)
+    procs.append(ServerProc(
+        [MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
+         f"http://127.0.0.1:{s3_port}", "--s3-bucket", "modelx", "--s3-access-key",
+         "modelx", "--s3-secret-key", "modelx123", "--enable-redirect"], mdx_port))
+    wait_http(mdx_port)
     if distributed:
         import torch.distributed as dist
 
@@ -208,6 +213,9 @@ def main():
             print(f"# stage {ph}: {secs:.2f}s {byts / (1 << 30):.2f} GiB "
                   f"{rate:.2f} GiB/s", file=sys.stderr)
 
+    for p in procs:
+        p.stop()
+    shutil.rmtree(store_root, ignore_errors=True)
     if rank == 0:
         moved_gib = 2.0 * args.blob_gib * args.steps * world  # push + pull, all ranks
         print(json.dumps({
@@ -226,9 +234,6 @@ def main():
             "config": {"model": f"synthetic-{args.blob_gib:g}GiB-blob",
                        "global_batch": world, "seq_len": 0,
                        "parallelism": f"dp{world}-presigned-s3"}}), flush=True)
-        for p in procs:
-            p.stop()
-        shutil.rmtree(store_root, ignore_errors=True)
     if distributed:
         import torch.distributed as dist
 
