@@ -1,0 +1,421 @@
+// zstd encoder core — standard RFC 8878 frames from greedy LZ77 matches,
+// raw literals and the predefined FSE sequence tables. Always decodable by
+// stock zstd; shared between the CDNA4 compress kernel (core/hip/zstd.hip)
+// and the CPU path (core/src/zstd_cpu.cpp).
+//
+// Same redundant-wavefront execution model as zstd_core.hpp: control flow
+// runs identically on all 64 lanes; literal byte moves go through
+// mx_par_copy.
+#pragma once
+
+#include "modelx/zstd_core.hpp"
+
+namespace modelx {
+namespace zstd {
+
+// ---------------------------------------------------------- bit writer -----
+
+// Forward byte emission, backward-readable (zstd bitstream convention):
+// bits accumulate LSB-first; close() appends the 1-marker then pads to a
+// byte boundary.
+struct BitW {
+  u8* dst;
+  u64 cap;
+  u64 pos;     // bytes written
+  u64 acc;
+  u32 nacc;    // bits in acc
+  bool overflow;
+
+  MX_HD void init(u8* d, u64 c) {
+    dst = d;
+    cap = c;
+    pos = 0;
+    acc = 0;
+    nacc = 0;
+    overflow = false;
+  }
+  MX_HD void add(u64 v, u32 nbits) {
+    if (nbits == 0) return;
+    acc |= (v & (((u64)1 << nbits) - 1)) << nacc;
+    nacc += nbits;
+    while (nacc >= 8) {
+      if (pos >= cap) {
+        overflow = true;
+        return;
+      }
+      dst[pos++] = (u8)acc;
+      acc >>= 8;
+      nacc -= 8;
+    }
+  }
+  MX_HD void close() {
+    add(1, 1);  // padding marker
+    if (nacc) {
+      if (pos >= cap) {
+        overflow = true;
+        return;
+      }
+      dst[pos++] = (u8)acc;
+      acc = 0;
+      nacc = 0;
+    }
+  }
+};
+
+// ------------------------------------------------------- FSE encoder -------
+
+struct FseEnc {
+  u16 state_table[1 << 6];   // predefined tables: log <= 6
+  i32 delta_find[53];
+  u32 delta_nbits[53];
+  u32 log;
+};
+
+MX_HD static inline int fse_build_ctable(FseEnc* e, const i16* counts, u32 nsym, u32 log) {
+  u32 size = 1u << log;
+  if (size > (1u << 6)) return MXZ_ERR_FSE;
+  u16 spread[1 << 6];
+  u32 high = size - 1;
+  for (u32 s = 0; s < nsym; s++)
+    if (counts[s] == -1) spread[high--] = (u16)s;
+  u32 step = (size >> 1) + (size >> 3) + 3;
+  u32 mask = size - 1;
+  u32 pos = 0;
+  for (u32 s = 0; s < nsym; s++) {
+    if (counts[s] <= 0) continue;
+    for (i32 i = 0; i < counts[s]; i++) {
+      spread[pos] = (u16)s;
+      do {
+        pos = (pos + step) & mask;
+      } while (pos > high);
+    }
+  }
+  if (pos != 0) return MXZ_ERR_FSE;
+  u32 cumul[54];
+  cumul[0] = 0;
+  for (u32 s = 0; s < nsym; s++)
+    cumul[s + 1] = cumul[s] + (u32)(counts[s] == -1 ? 1 : (counts[s] < 0 ? 0 : counts[s]));
+  u32 fill[54];
+  for (u32 s = 0; s <= nsym; s++) fill[s] = cumul[s];
+  for (u32 u = 0; u < size; u++) {
+    u32 s = spread[u];
+    e->state_table[fill[s]++] = (u16)(size + u);
+  }
+  u32 total = 0;
+  for (u32 s = 0; s < nsym; s++) {
+    i32 c = counts[s];
+    if (c == 0) {
+      e->delta_nbits[s] = ((log + 1) << 16) - size;
+      e->delta_find[s] = 0;
+    } else if (c == -1 || c == 1) {
+      e->delta_nbits[s] = (log << 16) - size;
+      e->delta_find[s] = (i32)total - 1;
+      total += 1;
+    } else {
+      u32 max_bits = log - mx_highbit((u32)c - 1);
+      e->delta_nbits[s] = (max_bits << 16) - ((u32)c << max_bits);
+      e->delta_find[s] = (i32)total - c;
+      total += (u32)c;
+    }
+  }
+  e->log = log;
+  return MXZ_OK;
+}
+
+struct FseState {
+  u32 value;
+};
+
+MX_HD static inline void fse_enc_init(const FseEnc* e, FseState* st, u32 sym) {
+  u32 nbits = (e->delta_nbits[sym] + (1u << 15)) >> 16;
+  u32 v = (nbits << 16) - e->delta_nbits[sym];
+  st->value = e->state_table[(v >> nbits) + (u32)((i32)e->delta_find[sym])];
+}
+
+MX_HD static inline void fse_enc_symbol(const FseEnc* e, FseState* st, BitW* bw, u32 sym) {
+  u32 nbits = (st->value + e->delta_nbits[sym]) >> 16;
+  bw->add(st->value, nbits);
+  st->value = e->state_table[(st->value >> nbits) + (u32)((i32)e->delta_find[sym])];
+}
+
+MX_HD static inline void fse_enc_flush(const FseEnc* e, FseState* st, BitW* bw) {
+  bw->add(st->value, e->log);
+}
+
+// ------------------------------------------------------ code mapping -------
+
+MX_HD static inline u32 ll_code_of(u32 ll) {
+  static const u8 kTab[64] = {0,  1,  2,  3,  4,  5,  6,  7,  8,  9,  10, 11, 12, 13, 14, 15,
+                              16, 16, 17, 17, 18, 18, 19, 19, 20, 20, 20, 20, 21, 21, 21, 21,
+                              22, 22, 22, 22, 22, 22, 22, 22, 23, 23, 23, 23, 23, 23, 23, 23,
+                              24, 24, 24, 24, 24, 24, 24, 24, 24, 24, 24, 24, 24, 24, 24, 24};
+  if (ll < 64) return kTab[ll];
+  u32 hb = mx_highbit(ll);
+  return hb + 19;  // 64..127 -> 25, 128.. -> 26, ... (2^h -> h+19)
+}
+
+MX_HD static inline u32 ml_code_of(u32 ml) {
+  // ml >= 3; mlBase = ml - 3
+  static const u8 kTab[128] = {
+      0,  1,  2,  3,  4,  5,  6,  7,  8,  9,  10, 11, 12, 13, 14, 15, 16, 17, 18, 19, 20, 21,
+      22, 23, 24, 25, 26, 27, 28, 29, 30, 31, 32, 32, 33, 33, 34, 34, 35, 35, 36, 36, 36, 36,
+      37, 37, 37, 37, 38, 38, 38, 38, 38, 38, 38, 38, 39, 39, 39, 39, 39, 39, 39, 39, 40, 40,
+      40, 40, 40, 40, 40, 40, 40, 40, 40, 40, 40, 40, 40, 40, 41, 41, 41, 41, 41, 41, 41, 41,
+      41, 41, 41, 41, 41, 41, 41, 41, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42,
+      42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42, 42};
+  u32 v = ml - 3;
+  if (v < 128) return kTab[v];
+  u32 hb = mx_highbit(v);
+  return hb + 36;  // 128.. -> 43, 256.. -> 44, ...
+}
+
+// ------------------------------------------------------------- LZ parse ----
+
+static const u32 kHashLog = 13;  // 8192-entry table, 32 KiB as u32
+
+MX_HD static inline u32 lz_hash(const u8* p) {
+  u32 v = mx_read_le32(p);
+  return (v * 2654435761u) >> (32 - kHashLog);
+}
+
+struct Seq {
+  u32 ll, ml, off;  // literal run length, match length, raw offset (>=1)
+};
+
+// Greedy parse of block[0..len) with matches back into base[0..block_end).
+// `hash` is a (1<<kHashLog) u32 table of position+1 into base, shared across
+// blocks of one frame (zeroed at frame start). Returns number of sequences.
+// seqs capacity must be >= len/4 + 1.
+MX_HD static inline u32 lz_parse(const u8* base, u64 block_off, u64 block_len, u32* hash,
+                                 Seq* seqs, u32 max_seqs, u64* lit_total) {
+  u64 end = block_off + block_len;
+  u64 p = block_off;
+  u64 lit_start = p;
+  u32 nseq = 0;
+  u64 limit = end >= 8 ? end - 8 : 0;  // room for u32 loads + extension
+  while (p < limit && nseq < max_seqs) {
+    u32 h = lz_hash(base + p);
+    u32 cand = hash[h];
+    hash[h] = (u32)(p + 1);
+    if (cand != 0) {
+      u64 cpos = cand - 1;
+      u64 dist = p - cpos;
+      if (dist > 0 && dist <= (u64)1 << 27 && mx_read_le32(base + cpos) == mx_read_le32(base + p)) {
+        // extend match
+        u64 m = 4;
+        while (p + m < end && base[cpos + m] == base[p + m]) m++;
+        seqs[nseq].ll = (u32)(p - lit_start);
+        seqs[nseq].ml = (u32)m;
+        seqs[nseq].off = (u32)dist;
+        nseq++;
+        // sparse hash inserts inside the match body
+        u64 stop = p + m < limit ? p + m : limit;
+        for (u64 q = p + 1; q < stop; q += 7) hash[lz_hash(base + q)] = (u32)(q + 1);
+        p += m;
+        lit_start = p;
+        continue;
+      }
+    }
+    p++;
+  }
+  u64 t = end - lit_start;  // trailing literal run
+  for (u32 i = 0; i < nseq; i++) t += seqs[i].ll;
+  *lit_total = t;
+  return nseq;
+}
+
+// ----------------------------------------------------------- block emit ----
+
+// Emit one zstd block (compressed if it wins, raw otherwise) for
+// src[block_off..block_off+block_len). Returns bytes written or <0.
+MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len, bool last,
+                                     u8* dst, u64 dstcap, u32* hash, Seq* seqs, u32 max_seqs,
+                                     const FseEnc* ell, const FseEnc* eof, const FseEnc* eml) {
+  const u8* block = src + block_off;
+  u64 raw_total = 3 + block_len;
+  if (dstcap < raw_total) return MXZ_ERR_DST_SMALL;
+
+  u64 lit_total = 0;
+  u32 nseq = lz_parse(src, block_off, block_len, hash, seqs, max_seqs, &lit_total);
+
+  bool use_raw = nseq == 0;
+  u64 csize = 0;
+  if (!use_raw) {
+    // literals header (raw literals)
+    u8* out = dst + 3;
+    u64 cap = dstcap - 3;
+    u64 lp = 0;
+    if (lit_total <= 31) {
+      out[lp++] = (u8)(0 | (0 << 2) | (lit_total << 3));
+    } else if (lit_total <= 4095) {
+      out[lp++] = (u8)(0 | (1 << 2) | ((lit_total & 0xF) << 4));
+      out[lp++] = (u8)(lit_total >> 4);
+    } else {
+      out[lp++] = (u8)(0 | (3 << 2) | ((lit_total & 0xF) << 4));
+      out[lp++] = (u8)((lit_total >> 4) & 0xFF);
+      out[lp++] = (u8)(lit_total >> 12);
+    }
+    if (lp + lit_total + 16 > cap) {
+      use_raw = true;
+    } else {
+      // literal bytes: replay the parse to copy each run
+      u64 p = block_off;
+      u64 w = lp;
+      for (u32 i = 0; i < nseq; i++) {
+        mx_par_copy(out + w, src + p, seqs[i].ll);
+        w += seqs[i].ll;
+        p += seqs[i].ll + seqs[i].ml;
+      }
+      u64 trail = (block_off + block_len) - p;
+      mx_par_copy(out + w, src + p, trail);
+      w += trail;
+      // nseq header
+      if (nseq < 128) {
+        out[w++] = (u8)nseq;
+      } else if (nseq < 0x7F00) {
+        out[w++] = (u8)((nseq >> 8) + 128);
+        out[w++] = (u8)(nseq & 0xFF);
+      } else {
+        out[w++] = 255;
+        out[w++] = (u8)((nseq - 0x7F00) & 0xFF);
+        out[w++] = (u8)((nseq - 0x7F00) >> 8);
+      }
+      out[w++] = 0;  // modes: all predefined
+      // sequence bitstream (encode in reverse)
+      BitW bw;
+      bw.init(out + w, cap - w > block_len ? block_len : cap - w);  // bounded: must beat raw
+      const Seq& lastq = seqs[nseq - 1];
+      u32 ll_c = ll_code_of(lastq.ll);
+      u32 ml_c = ml_code_of(lastq.ml);
+      u64 of_v = (u64)lastq.off + 3;
+      u32 of_c = mx_highbit((u32)of_v);
+      FseState sll, sof, sml;
+      fse_enc_init(eml, &sml, ml_c);
+      fse_enc_init(eof, &sof, of_c);
+      fse_enc_init(ell, &sll, ll_c);
+      u32 llb, llbase, mlb, mlbase;
+      ll_code_info(ll_c, &llb, &llbase);
+      ml_code_info(ml_c, &mlb, &mlbase);
+      bw.add(lastq.ll - llbase, llb);
+      bw.add(lastq.ml - mlbase, mlb);
+      bw.add(of_v - ((u64)1 << of_c), of_c);
+      for (i32 i = (i32)nseq - 2; i >= 0; i--) {
+        const Seq& q = seqs[i];
+        u32 lc = ll_code_of(q.ll);
+        u32 mc = ml_code_of(q.ml);
+        u64 ov = (u64)q.off + 3;
+        u32 oc = mx_highbit((u32)ov);
+        fse_enc_symbol(eof, &sof, &bw, oc);
+        fse_enc_symbol(eml, &sml, &bw, mc);
+        fse_enc_symbol(ell, &sll, &bw, lc);
+        ll_code_info(lc, &llb, &llbase);
+        ml_code_info(mc, &mlb, &mlbase);
+        bw.add(q.ll - llbase, llb);
+        bw.add(q.ml - mlbase, mlb);
+        bw.add(ov - ((u64)1 << oc), oc);
+        if (bw.overflow) break;
+      }
+      fse_enc_flush(eml, &sml, &bw);
+      fse_enc_flush(eof, &sof, &bw);
+      fse_enc_flush(ell, &sll, &bw);
+      bw.close();
+      if (bw.overflow) {
+        use_raw = true;
+      } else {
+        csize = w + bw.pos;
+        if (csize >= block_len) use_raw = true;
+      }
+    }
+  }
+
+  if (use_raw) {
+    u32 bh = ((u32)block_len << 3) | (0 << 1) | (last ? 1 : 0);
+    dst[0] = (u8)bh;
+    dst[1] = (u8)(bh >> 8);
+    dst[2] = (u8)(bh >> 16);
+    mx_par_copy(dst + 3, block, block_len);
+    return (i64)(3 + block_len);
+  }
+  u32 bh = ((u32)csize << 3) | (2 << 1) | (last ? 1 : 0);
+  dst[0] = (u8)bh;
+  dst[1] = (u8)(bh >> 8);
+  dst[2] = (u8)(bh >> 16);
+  return (i64)(3 + csize);
+}
+
+// Encode src[0..len) as ONE standard zstd frame into dst. `hash` is a
+// (1<<kHashLog) u32 scratch (will be zeroed), `seqs` holds >= kBlockMax/4+1
+// entries. Returns frame size or <0.
+MX_HD static inline i64 encode_frame(const u8* src, u64 len, u8* dst, u64 dstcap, u32* hash,
+                                     Seq* seqs) {
+  FseEnc ell, eof, eml;
+  u32 nsym, log;
+  const i16* d = ll_default_dist(&nsym, &log);
+  if (fse_build_ctable(&ell, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  d = of_default_dist(&nsym, &log);
+  if (fse_build_ctable(&eof, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  d = ml_default_dist(&nsym, &log);
+  if (fse_build_ctable(&eml, d, nsym, log) < 0) return MXZ_ERR_FSE;
+
+  {
+    u32 lane = mx_lane(), w = mx_width();
+    for (u32 i = lane; i < (1u << kHashLog); i += w) hash[i] = 0;
+    mx_sync();
+  }
+
+  u64 pos = 0;
+  if (dstcap < 16) return MXZ_ERR_DST_SMALL;
+  mx_write_le32(dst, kMagic);
+  pos = 4;
+  // frame header: single-segment, FCS sized to len
+  u32 fcs_flag;
+  if (len < 256)
+    fcs_flag = 0;
+  else if (len < 65536 + 256)
+    fcs_flag = 1;
+  else if (len <= 0xFFFFFFFFull)
+    fcs_flag = 2;
+  else
+    fcs_flag = 3;
+  dst[pos++] = (u8)((fcs_flag << 6) | (1u << 5));
+  switch (fcs_flag) {
+    case 0:
+      dst[pos++] = (u8)len;
+      break;
+    case 1: {
+      u64 v = len - 256;
+      dst[pos++] = (u8)v;
+      dst[pos++] = (u8)(v >> 8);
+      break;
+    }
+    case 2:
+      mx_write_le32(dst + pos, (u32)len);
+      pos += 4;
+      break;
+    default:
+      for (u32 i = 0; i < 8; i++) dst[pos++] = (u8)(len >> (8 * i));
+      break;
+  }
+  if (len == 0) {
+    // single empty raw block
+    if (pos + 3 > dstcap) return MXZ_ERR_DST_SMALL;
+    dst[pos] = 1;  // last=1, type=raw, size=0
+    dst[pos + 1] = 0;
+    dst[pos + 2] = 0;
+    return (i64)(pos + 3);
+  }
+  u64 off = 0;
+  while (off < len) {
+    u64 blen = len - off < kBlockMax ? len - off : kBlockMax;
+    bool last = off + blen >= len;
+    i64 n = encode_block(src, off, blen, last, dst + pos, dstcap - pos, hash, seqs,
+                         (u32)(kBlockMax / 4 + 1), &ell, &eof, &eml);
+    if (n < 0) return n;
+    pos += (u64)n;
+    off += blen;
+  }
+  return (i64)pos;
+}
+
+}  // namespace zstd
+}  // namespace modelx
