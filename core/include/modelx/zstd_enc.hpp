@@ -343,19 +343,30 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
   return (i64)(3 + csize);
 }
 
-// Encode src[0..len) as ONE standard zstd frame into dst. `hash` is a
-// (1<<kHashLog) u32 scratch (will be zeroed), `seqs` holds >= kBlockMax/4+1
-// entries. Returns frame size or <0.
-MX_HD static inline i64 encode_frame(const u8* src, u64 len, u8* dst, u64 dstcap, u32* hash,
-                                     Seq* seqs) {
+// Predefined-table FSE encoders (caller-allocated: LDS on the GPU so the
+// redundant-wavefront execution doesn't spill per-lane copies).
+struct EncTables {
   FseEnc ell, eof, eml;
+};
+
+MX_HD static inline int enc_tables_init(EncTables* et) {
   u32 nsym, log;
   const i16* d = ll_default_dist(&nsym, &log);
-  if (fse_build_ctable(&ell, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  if (fse_build_ctable(&et->ell, d, nsym, log) < 0) return MXZ_ERR_FSE;
   d = of_default_dist(&nsym, &log);
-  if (fse_build_ctable(&eof, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  if (fse_build_ctable(&et->eof, d, nsym, log) < 0) return MXZ_ERR_FSE;
   d = ml_default_dist(&nsym, &log);
-  if (fse_build_ctable(&eml, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  if (fse_build_ctable(&et->eml, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  return MXZ_OK;
+}
+
+// Encode src[0..len) as ONE standard zstd frame into dst. `hash` is a
+// (1<<kHashLog) u32 scratch (will be zeroed), `seqs` holds >= kBlockMax/4+1
+// entries, `et` holds initialized predefined tables (enc_tables_init).
+// Returns frame size or <0.
+MX_HD static inline i64 encode_frame(const u8* src, u64 len, u8* dst, u64 dstcap, u32* hash,
+                                     Seq* seqs, const EncTables* et) {
+  const FseEnc &ell = et->ell, &eof = et->eof, &eml = et->eml;
 
   {
     u32 lane = mx_lane(), w = mx_width();

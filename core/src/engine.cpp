@@ -37,6 +37,8 @@
 #include <vector>
 
 #include "modelx/http.hpp"
+#include "modelx/zstd_core.hpp"
+#include "modelx/zstd_host.hpp"
 
 namespace py = pybind11;
 using namespace modelx;
@@ -51,6 +53,20 @@ extern "C" hipError_t modelx_tar_index(const void* tar, uint64_t tar_len, void* 
                                        uint32_t* error_dev, hipStream_t stream);
 extern "C" hipError_t modelx_tar_scatter(const void* tar, const void* segs, uint32_t nsegs,
                                          hipStream_t stream);
+extern "C" hipError_t modelx_zstd_decompress_frames(const void* src, const void* frames_dev,
+                                                    uint32_t nframes, void* dst,
+                                                    void* lit_scratch, int64_t* rc_dev,
+                                                    hipStream_t stream);
+extern "C" hipError_t modelx_zstd_compress_frames(const void* src, uint64_t srclen,
+                                                  uint32_t frame_raw, uint32_t first_frame,
+                                                  uint32_t nframes, void* dst_scratch,
+                                                  uint64_t stride, void* seq_scratch,
+                                                  uint32_t max_seqs, int64_t* out_sizes_dev,
+                                                  hipStream_t stream);
+
+struct MxzFrameHost {
+  uint64_t c_off, c_size, d_off, d_size;
+};
 
 struct TarEntryHost {
   uint64_t header_off, payload_off, size;
@@ -538,6 +554,149 @@ class GpuEngine {
     hipFree(dsegs);
   }
 
+  // ------------------------------------------------------------- zstd ----
+
+  // Compress a device buffer into a seekable multi-frame zstd blob at
+  // dst_ptr (device). Returns the total blob size (frames + seek table).
+  // One workgroup per frame (core/hip/zstd.hip); frames are compressed into
+  // padded per-frame scratch in batches, packed with the scatter kernel,
+  // and the seek table is appended from the host.
+  uint64_t zstd_compress_device(uintptr_t src_ptr, uint64_t size, uint32_t frame_raw,
+                                uintptr_t dst_ptr, uint64_t dst_cap) {
+    HIP_CHECK(hipSetDevice(device_));
+    if (frame_raw == 0) frame_raw = 128 << 10;
+    using namespace modelx::zstd;
+    uint64_t nframes = size ? (size + frame_raw - 1) / frame_raw : 1;
+    uint64_t stride = (uint64_t)frame_raw + 64 + 3 * ((frame_raw + kBlockMax - 1) / kBlockMax);
+    uint32_t max_seqs = kBlockMax / 4 + 1;
+    uint64_t batch = std::min<uint64_t>(nframes, 1024);
+    std::vector<zstdhost::SeekEntry> entries(nframes);
+    py::gil_scoped_release release;
+    void* dscratch = nullptr;
+    void* dseqs = nullptr;
+    int64_t* dsizes = nullptr;
+    void* dsegs = nullptr;
+    HIP_CHECK(hipMalloc(&dscratch, batch * stride));
+    HIP_CHECK(hipMalloc(&dseqs, batch * (uint64_t)max_seqs * 12));
+    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dsizes), batch * sizeof(int64_t)));
+    HIP_CHECK(hipMalloc(&dsegs, batch * sizeof(CopySegHost)));
+    std::vector<int64_t> hsizes(batch);
+    std::vector<CopySegHost> hsegs(batch);
+    uint64_t out = 0;
+    for (uint64_t first = 0; first < nframes; first += batch) {
+      uint32_t n = (uint32_t)std::min<uint64_t>(batch, nframes - first);
+      HIP_CHECK(modelx_zstd_compress_frames(reinterpret_cast<void*>(src_ptr), size, frame_raw,
+                                            (uint32_t)first, n, dscratch, stride, dseqs,
+                                            max_seqs, dsizes, hash_stream_));
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+      HIP_CHECK(hipMemcpy(hsizes.data(), dsizes, n * sizeof(int64_t), hipMemcpyDeviceToHost));
+      for (uint32_t i = 0; i < n; i++) {
+        if (hsizes[i] < 0)
+          throw std::runtime_error("zstd compress kernel failed: frame " +
+                                   std::to_string(first + i) + " rc=" +
+                                   std::to_string(hsizes[i]));
+        uint64_t f = first + i;
+        uint64_t d_off = f * (uint64_t)frame_raw;
+        entries[f] = {out, (uint64_t)hsizes[i], d_off,
+                      std::min<uint64_t>(frame_raw, size - std::min(size, d_off))};
+        if (size == 0) entries[f].d_size = 0;
+        hsegs[i] = {i * stride, dst_ptr + out, (uint64_t)hsizes[i]};
+        out += (uint64_t)hsizes[i];
+        if (out > dst_cap) throw std::runtime_error("zstd compress: dst overflow");
+      }
+      HIP_CHECK(hipMemcpy(dsegs, hsegs.data(), n * sizeof(CopySegHost), hipMemcpyHostToDevice));
+      HIP_CHECK(modelx_tar_scatter(dscratch, dsegs, n, hash_stream_));
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+    }
+    auto table = zstdhost::build_seek_table(entries);
+    if (out + table.size() > dst_cap) throw std::runtime_error("zstd compress: dst overflow");
+    HIP_CHECK(hipMemcpy(reinterpret_cast<char*>(dst_ptr) + out, table.data(), table.size(),
+                        hipMemcpyHostToDevice));
+    out += table.size();
+    hipFree(dscratch);
+    hipFree(dseqs);
+    hipFree(dsizes);
+    hipFree(dsegs);
+    return out;
+  }
+
+  // Decompress a seekable multi-frame zstd blob resident in HBM into
+  // dst_ptr. Parses the seek table from the blob tail (one small D2H copy),
+  // then decodes every frame in its own workgroup. Returns decompressed
+  // size.
+  uint64_t zstd_decompress_device(uintptr_t src_ptr, uint64_t src_len, uintptr_t dst_ptr,
+                                  uint64_t dst_cap) {
+    HIP_CHECK(hipSetDevice(device_));
+    using namespace modelx::zstd;
+    py::gil_scoped_release release;
+    // read the seek-table footer
+    if (src_len < 17) throw std::runtime_error("zstd blob too small");
+    uint8_t foot[17];
+    HIP_CHECK(hipMemcpy(foot, reinterpret_cast<char*>(src_ptr) + src_len - 17, 17,
+                        hipMemcpyDeviceToHost));
+    uint32_t magic = (uint32_t)foot[13] | ((uint32_t)foot[14] << 8) | ((uint32_t)foot[15] << 16) |
+                     ((uint32_t)foot[16] << 24);
+    if (magic != kSeekTableMagic)
+      throw std::runtime_error("zstd blob has no seek table (footer magic mismatch)");
+    bool checksums = foot[12] & 0x80;
+    uint32_t nframes = (uint32_t)foot[8] | ((uint32_t)foot[9] << 8) | ((uint32_t)foot[10] << 16) |
+                       ((uint32_t)foot[11] << 24);
+    uint64_t entry_sz = checksums ? 12 : 8;
+    uint64_t tbl = 8 + (uint64_t)nframes * entry_sz + 9;
+    if (tbl > src_len) throw std::runtime_error("zstd seek table larger than blob");
+    std::vector<uint8_t> traw(tbl);
+    HIP_CHECK(hipMemcpy(traw.data(), reinterpret_cast<char*>(src_ptr) + src_len - tbl, tbl,
+                        hipMemcpyDeviceToHost));
+    std::vector<MxzFrameHost> frames(nframes);
+    uint64_t c_off = 0, d_off = 0;
+    for (uint32_t i = 0; i < nframes; i++) {
+      const uint8_t* e = traw.data() + 8 + (uint64_t)i * entry_sz;
+      uint32_t cs = (uint32_t)e[0] | ((uint32_t)e[1] << 8) | ((uint32_t)e[2] << 16) |
+                    ((uint32_t)e[3] << 24);
+      uint32_t ds = (uint32_t)e[4] | ((uint32_t)e[5] << 8) | ((uint32_t)e[6] << 16) |
+                    ((uint32_t)e[7] << 24);
+      frames[i] = {c_off, cs, d_off, ds};
+      c_off += cs;
+      d_off += ds;
+    }
+    if (d_off > dst_cap) throw std::runtime_error("zstd decompress: dst too small");
+    uint64_t batch = std::min<uint64_t>(nframes ? nframes : 1, 8192);
+    void* dframes = nullptr;
+    void* dlit = nullptr;
+    int64_t* drc = nullptr;
+    HIP_CHECK(hipMalloc(&dframes, batch * sizeof(MxzFrameHost)));
+    HIP_CHECK(hipMalloc(&dlit, batch * (uint64_t)kBlockMax));
+    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&drc), batch * sizeof(int64_t)));
+    std::vector<int64_t> hrc(batch);
+    for (uint64_t first = 0; first < nframes; first += batch) {
+      uint32_t n = (uint32_t)std::min<uint64_t>(batch, nframes - first);
+      HIP_CHECK(hipMemcpy(dframes, frames.data() + first, n * sizeof(MxzFrameHost),
+                          hipMemcpyHostToDevice));
+      HIP_CHECK(modelx_zstd_decompress_frames(reinterpret_cast<void*>(src_ptr), dframes, n,
+                                              reinterpret_cast<void*>(dst_ptr), dlit, drc,
+                                              hash_stream_));
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+      HIP_CHECK(hipMemcpy(hrc.data(), drc, n * sizeof(int64_t), hipMemcpyDeviceToHost));
+      for (uint32_t i = 0; i < n; i++)
+        if (hrc[i] != 0)
+          throw std::runtime_error("zstd decompress kernel failed: frame " +
+                                   std::to_string(first + i) + " rc=" + std::to_string(hrc[i]));
+    }
+    hipFree(dframes);
+    hipFree(dlit);
+    hipFree(drc);
+    return d_off;
+  }
+
+  // Upper bound for zstd_compress_device output.
+  static uint64_t zstd_compress_bound(uint64_t size, uint32_t frame_raw) {
+    using namespace modelx::zstd;
+    if (frame_raw == 0) frame_raw = 128 << 10;
+    uint64_t nframes = size ? (size + frame_raw - 1) / frame_raw : 1;
+    uint64_t stride = (uint64_t)frame_raw + 64 + 3 * ((frame_raw + kBlockMax - 1) / kBlockMax);
+    return nframes * stride + 8 + nframes * 8 + 9;
+  }
+
   void synchronize() {
     HIP_CHECK(hipSetDevice(device_));
     for (auto& st : streams_) HIP_CHECK(hipStreamSynchronize(st));
@@ -626,5 +785,49 @@ PYBIND11_MODULE(_core, m) {
            py::arg("method"), py::arg("headers"), py::arg("src_ptr"), py::arg("size"))
       .def("tar_index", &GpuEngine::tar_index, py::arg("tar_ptr"), py::arg("tar_len"))
       .def("tar_scatter", &GpuEngine::tar_scatter, py::arg("tar_ptr"), py::arg("segs"))
+      .def("zstd_compress_device", &GpuEngine::zstd_compress_device, py::arg("src_ptr"),
+           py::arg("size"), py::arg("frame_raw"), py::arg("dst_ptr"), py::arg("dst_cap"))
+      .def("zstd_decompress_device", &GpuEngine::zstd_decompress_device, py::arg("src_ptr"),
+           py::arg("src_len"), py::arg("dst_ptr"), py::arg("dst_cap"))
       .def("synchronize", &GpuEngine::synchronize);
+
+  m.def("zstd_compress_bound", &GpuEngine::zstd_compress_bound, py::arg("size"),
+        py::arg("frame_raw") = (uint32_t)(128 << 10));
+  // CPU codec path (same header-only core as the kernels) — used by the
+  // CPU client for +zstd blobs and by tests as the GPU-vs-CPU oracle.
+  m.def(
+      "zstd_compress_cpu",
+      [](py::bytes data, uint32_t frame_raw) {
+        std::string s = data;
+        std::vector<uint8_t> out;
+        {
+          py::gil_scoped_release release;
+          out = zstdhost::compress_seekable(reinterpret_cast<const uint8_t*>(s.data()),
+                                            s.size(), frame_raw);
+        }
+        return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
+      },
+      py::arg("data"), py::arg("frame_raw") = (uint32_t)(128 << 10));
+  m.def("zstd_decompress_cpu", [](py::bytes data) {
+    std::string s = data;
+    std::vector<uint8_t> out;
+    {
+      py::gil_scoped_release release;
+      out = zstdhost::decompress(reinterpret_cast<const uint8_t*>(s.data()), s.size());
+    }
+    return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
+  });
+  m.def("zstd_content_size", [](py::bytes data) {
+    std::string s = data;
+    return zstdhost::content_size(reinterpret_cast<const uint8_t*>(s.data()), s.size());
+  });
+  m.def("zstd_frames", [](py::bytes data) {
+    std::string s = data;
+    auto t = zstdhost::parse_seek_table(reinterpret_cast<const uint8_t*>(s.data()), s.size());
+    if (t.empty())
+      t = zstdhost::walk_frames(reinterpret_cast<const uint8_t*>(s.data()), s.size());
+    py::list out;
+    for (auto& e : t) out.append(py::make_tuple(e.c_off, e.c_size, e.d_off, e.d_size));
+    return out;
+  });
 }

@@ -175,12 +175,14 @@ std::vector<uint8_t> compress_seekable(const uint8_t* src, size_t len, uint32_t 
     pool.emplace_back([&] {
       std::vector<u32> hash(1u << kHashLog);
       std::vector<Seq> seqs(kBlockMax / 4 + 1);
+      EncTables et;
+      enc_tables_init(&et);
       size_t i;
       while ((i = next.fetch_add(1)) < nframes) {
         uint64_t off = (uint64_t)i * frame_raw;
         uint64_t flen = len - off < frame_raw ? len - off : frame_raw;
         sizes[i] = encode_frame(src + off, flen, scratch.data() + i * stride, stride,
-                                hash.data(), seqs.data());
+                                hash.data(), seqs.data(), &et);
       }
     });
   }

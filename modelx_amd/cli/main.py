@@ -59,7 +59,7 @@ def cmd_push(args) -> int:
     if not ref.version:
         ref.version = "latest"
     manifest = client.push(ref.repository, ref.version, args.dir, configfile=args.config,
-                           digest_mode=args.digest_mode)
+                           digest_mode=args.digest_mode, compress=args.compress)
     total = manifest.config.size + sum(b.size for b in manifest.blobs)
     print(f"pushed {ref.repository}@{ref.version}: {len(manifest.blobs)} blobs, "
           f"{human_size(total)}")
@@ -209,6 +209,9 @@ def build_parser() -> argparse.ArgumentParser:
     sp.add_argument("dir", nargs="?", default=".")
     sp.add_argument("--config", default=MODEL_CONFIG_FILENAME)
     sp.add_argument("--digest-mode", choices=["sha256", "chunked"], default="sha256")
+    sp.add_argument("--compress", choices=["", "zstd"], default="",
+                    help="store file blobs zstd-compressed (seekable multi-frame, "
+                         "GPU-parallel decode)")
     sp.set_defaults(fn=cmd_push)
 
     sp = sub.add_parser("pull", help="pull a model version")

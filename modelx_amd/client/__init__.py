@@ -23,9 +23,11 @@ class Client:
 
     def push(self, repository: str, version: str, basedir: str,
              configfile: str = "modelx.yaml", digest_mode: str = "sha256",
-             quiet: Optional[bool] = None, dir_format: str = "tar+gz") -> types.Manifest:
+             quiet: Optional[bool] = None, dir_format: str = "tar+gz",
+             compress: str = "") -> types.Manifest:
         return self.pusher.push(repository, version or "latest", basedir, configfile,
-                                digest_mode=digest_mode, quiet=quiet, dir_format=dir_format)
+                                digest_mode=digest_mode, quiet=quiet, dir_format=dir_format,
+                                compress=compress)
 
     def pull(self, repository: str, version: str, into_dir: str,
              quiet: Optional[bool] = None) -> types.Manifest:

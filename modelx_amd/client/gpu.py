@@ -154,6 +154,42 @@ class GpuClient:
             total += ln
         return total
 
+    def pull_zstd_blob_to_device(self, repository: str, desc: types.Descriptor,
+                                 verify: bool = True) -> "torch.Tensor":
+        """Pull a +zstd blob: land the compressed stream in HBM (streaming
+        SHA-256 verify over the stored bytes), decode every frame in its own
+        workgroup (core/hip/zstd.hip), then GPU-verify the uncompressed
+        chunked digest from the raw-digest annotation. The whole path —
+        transfer, hash, inflate, re-hash — never leaves the GPU."""
+        import time
+
+        import torch
+
+        comp = self.pull_blob_to_device(repository, desc, verify=verify)
+        raw_size = int(desc.annotations.get(types.ANNOTATION_RAW_SIZE, 0) or 0)
+        if not raw_size:
+            raise er.ModelxError(er.ErrCode.UNSUPPORTED,
+                                 f"+zstd blob {desc.name} lacks the raw-size annotation")
+        out = torch.empty(max(raw_size, 1), dtype=torch.uint8, device=f"cuda:{self.device}")
+        t0 = time.monotonic()
+        got = self.engine.zstd_decompress_device(comp.data_ptr(), desc.size,
+                                                 out.data_ptr(), out.numel())
+        self.last_stats.append({"phase": "pull-zstd-decompress", "bytes": got,
+                                "seconds": time.monotonic() - t0})
+        if raw_size and got != raw_size:
+            raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
+                                 f"zstd size mismatch for {desc.name}: {got} != {raw_size}")
+        out = out[:got]
+        raw_digest = desc.annotations.get(types.ANNOTATION_RAW_DIGEST, "")
+        if verify and raw_digest:
+            cs = dg.algo_chunk_size(raw_digest.split(":", 1)[0]) or DEFAULT_GPU_CHUNK
+            leaves = self.engine.sha256_chunk_leaves(out.data_ptr(), got, cs)
+            root = dg.root_from_leaf_bytes(leaves, cs, got)
+            if root != raw_digest:
+                raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
+                                     f"uncompressed digest mismatch for {desc.name}")
+        return out
+
     def pull_blob_to_device(self, repository: str, desc: types.Descriptor,
                             tensor=None, verify: bool = True,
                             resume: bool = False) -> "torch.Tensor":
@@ -248,7 +284,10 @@ class GpuClient:
         for desc in manifest.blobs:
             if desc.size == 0 or desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
                 continue
-            out[desc.name] = self.pull_blob_to_device(repository, desc, verify=verify)
+            if desc.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
+                out[desc.name] = self.pull_zstd_blob_to_device(repository, desc, verify=verify)
+            else:
+                out[desc.name] = self.pull_blob_to_device(repository, desc, verify=verify)
         return out
 
     # -------------------------------------------------- directory blobs --
@@ -367,8 +406,17 @@ class GpuClient:
     def push_from_gpu(self, repository: str, version: str,
                       tensors: Dict[str, "torch.Tensor"], config_yaml: str = "",
                       chunk_size: int = DEFAULT_GPU_CHUNK,
-                      part_bytes: int = DEFAULT_PART_BYTES) -> types.Manifest:
-        """Digest on GPU → presigned multipart upload → manifest PUT last."""
+                      part_bytes: int = DEFAULT_PART_BYTES,
+                      compress: str = "") -> types.Manifest:
+        """Digest on GPU → presigned multipart upload → manifest PUT last.
+        With ``compress="zstd"`` each tensor is compressed on-GPU into a
+        seekable multi-frame zstd blob (core/hip/zstd.hip) first; the
+        descriptor digest covers the stored (compressed) bytes and the
+        raw-digest/raw-size annotations carry the uncompressed identity."""
+        import torch
+
+        if compress not in ("", "zstd"):
+            raise er.ModelxError(er.ErrCode.UNSUPPORTED, f"unknown compression {compress!r}")
         manifest = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
         now = datetime.now(timezone.utc)
         # config blob (small, CPU)
@@ -380,16 +428,42 @@ class GpuClient:
         if not self.remote.head_blob(repository, cfg_digest):
             self._upload_small(repository, manifest.config, cfg)
         for name, t in tensors.items():
+            import time
+
             size = t.numel() * t.element_size()
-            root, leaves = self.digest_device_blob_with_leaves(t.data_ptr(), size, chunk_size)
+            media_type = types.MEDIA_TYPE_MODEL_FILE
+            extra_notes: Dict[str, str] = {}
+            push_ptr, push_size = t.data_ptr(), size
+            comp_keep = None  # keep the compressed tensor alive until pushed
+            if compress == "zstd" and size:
+                core = _core()
+                raw_root, _ = self.digest_device_blob_with_leaves(t.data_ptr(), size,
+                                                                  chunk_size)
+                bound = core.zstd_compress_bound(size)
+                comp_keep = torch.empty(bound, dtype=torch.uint8,
+                                        device=f"cuda:{self.device}")
+                t0 = time.monotonic()
+                comp_size = self.engine.zstd_compress_device(
+                    t.data_ptr(), size, 128 << 10, comp_keep.data_ptr(), bound)
+                self.last_stats.append({"phase": "push-zstd-compress", "bytes": size,
+                                        "seconds": time.monotonic() - t0,
+                                        "compressed": comp_size})
+                media_type = types.MEDIA_TYPE_MODEL_FILE_ZSTD
+                extra_notes = {types.ANNOTATION_RAW_DIGEST: raw_root,
+                               types.ANNOTATION_RAW_SIZE: str(size)}
+                push_ptr, push_size = comp_keep.data_ptr(), comp_size
+            root, leaves = self.digest_device_blob_with_leaves(push_ptr, push_size,
+                                                               chunk_size)
             leaves_digest = dg.sha256_digest(leaves)
             desc = types.Descriptor(
-                name=name, media_type=types.MEDIA_TYPE_MODEL_FILE, digest=root, size=size,
+                name=name, media_type=media_type, digest=root, size=push_size,
                 modified=now,
                 annotations={types.ANNOTATION_CHUNK_DIGEST: root,
                              types.ANNOTATION_CHUNK_SIZE: str(chunk_size),
-                             types.ANNOTATION_LEAVES_BLOB: leaves_digest})
-            self.push_blob_from_device(repository, desc, t.data_ptr(), part_bytes=part_bytes)
+                             types.ANNOTATION_LEAVES_BLOB: leaves_digest,
+                             **extra_notes})
+            self.push_blob_from_device(repository, desc, push_ptr, part_bytes=part_bytes)
+            del comp_keep
             manifest.blobs.append(desc)
             # leaves sidecar: 32 B per chunk, enables chunk-level
             # resume/refetch/dedup on pull; listed in the manifest so GC

@@ -62,6 +62,8 @@ class Puller:
         elif desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
             if bar:
                 bar.set_status("sidecar", complete=True)
+        elif desc.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
+            self._pull_zstd_file(repository, desc, into_dir, bar)
         else:
             self._pull_file(repository, desc, into_dir, bar)
 
@@ -78,6 +80,37 @@ class Puller:
         self.pull_blob(repository, desc, dest, bar)
         if desc.mode:
             os.chmod(dest, desc.mode & 0o7777)
+
+    def _pull_zstd_file(self, repository: str, desc: types.Descriptor, into_dir: str,
+                        bar: Optional[Bar]) -> None:
+        """CPU pull of a +zstd blob (GPU path: GpuClient.pull_zstd_blob_to_device).
+        The stored bytes are verified by pull_blob against desc.digest; the
+        decompressed output is verified against the raw-digest annotation."""
+        from modelx_amd import _core
+        from ..wire import digest as dg
+
+        dest = os.path.join(into_dir, desc.name)
+        raw_digest = desc.annotations.get(types.ANNOTATION_RAW_DIGEST, "")
+        if os.path.isfile(dest) and raw_digest and _verify_digest_of_file(dest, raw_digest):
+            if bar:
+                bar.set_status("up to date", complete=True)
+            return
+        cache_dir = os.path.join(into_dir, MODELX_CACHE_DIR)
+        os.makedirs(cache_dir, exist_ok=True)
+        comp_path = os.path.join(cache_dir, desc.name + ".zst")
+        if not (os.path.isfile(comp_path) and _verify_digest_of_file(comp_path, desc.digest)):
+            self.pull_blob(repository, desc, comp_path, bar)
+        with open(comp_path, "rb") as f:
+            raw = _core.zstd_decompress_cpu(f.read())
+        if raw_digest and not dg.verify_bytes(raw, raw_digest):
+            raise ValueError(f"uncompressed digest mismatch for {desc.name}")
+        tmp = dest + ".part"
+        with open(tmp, "wb") as f:
+            f.write(raw)
+        os.replace(tmp, dest)
+        if desc.mode:
+            os.chmod(dest, desc.mode & 0o7777)
+        os.remove(comp_path)
 
     # ------------------------------------------------------- directories --
 

@@ -51,12 +51,46 @@ def _file_descriptor(path: str, name: str, chunk_size: int, digest_mode: str) ->
     return desc
 
 
+def _zstd_file_descriptor(path: str, name: str, chunk_size: int,
+                          digest_mode: str, cache_dir: str) -> types.Descriptor:
+    """Compress a file into the seekable multi-frame zstd format (CPU path
+    of core/hip/zstd.hip's codec) and describe the COMPRESSED blob; the
+    uncompressed identity travels in raw-digest/raw-size annotations."""
+    from modelx_amd import _core
+
+    st = os.stat(path)
+    with open(path, "rb") as f:
+        raw = f.read()
+    blob = _core.zstd_compress_cpu(raw, 128 << 10)
+    comp_path = os.path.join(cache_dir, name + ".zst")
+    with open(comp_path, "wb") as f:
+        f.write(blob)
+    raw_chunked = dg.chunked_digest(raw, chunk_size)
+    canonical = dg.sha256_digest(blob)
+    chunked = dg.chunked_digest(blob, chunk_size)
+    return types.Descriptor(
+        name=name,
+        media_type=types.MEDIA_TYPE_MODEL_FILE_ZSTD,
+        digest=chunked if digest_mode == "chunked" else canonical,
+        size=len(blob),
+        mode=stat.S_IMODE(st.st_mode),
+        modified=datetime.fromtimestamp(st.st_mtime, tz=timezone.utc),
+        annotations={
+            types.ANNOTATION_CHUNK_DIGEST: chunked,
+            types.ANNOTATION_CHUNK_SIZE: str(chunk_size),
+            types.ANNOTATION_RAW_DIGEST: raw_chunked,
+            types.ANNOTATION_RAW_SIZE: str(len(raw)),
+        },
+    )
+
+
 def parse_manifest(basedir: str, configfile: str = "modelx.yaml",
                    chunk_size: int = dg.DEFAULT_CHUNK_SIZE,
                    digest_mode: str = "sha256",
-                   dir_format: str = "tar+gz") -> types.Manifest:
+                   dir_format: str = "tar+gz", compress: str = "") -> types.Manifest:
     """Scan basedir into a Manifest (reference: push.go:67-100).
-    Dot-files skipped; directories become tar.gz descriptors."""
+    Dot-files skipped; directories become tar.gz descriptors.
+    compress="zstd" stores file blobs in the seekable +zstd format."""
     manifest = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
     cache_dir = os.path.join(basedir, MODELX_CACHE_DIR)
     os.makedirs(cache_dir, exist_ok=True)
@@ -91,6 +125,9 @@ def parse_manifest(basedir: str, configfile: str = "modelx.yaml",
                 },
             )
             manifest.blobs.append(desc)
+        elif compress == "zstd":
+            manifest.blobs.append(
+                _zstd_file_descriptor(path, entry, chunk_size, digest_mode, cache_dir))
         else:
             manifest.blobs.append(_file_descriptor(path, entry, chunk_size, digest_mode))
     manifest.blobs = types.sort_descriptors_by_name(manifest.blobs)  # push.go:98
@@ -107,8 +144,9 @@ class Pusher:
     def push(self, repository: str, version: str, basedir: str,
              configfile: str = "modelx.yaml", digest_mode: str = "sha256",
              chunk_size: int = dg.DEFAULT_CHUNK_SIZE, quiet: Optional[bool] = None,
-             dir_format: str = "tar+gz") -> types.Manifest:
-        manifest = parse_manifest(basedir, configfile, chunk_size, digest_mode, dir_format)
+             dir_format: str = "tar+gz", compress: str = "") -> types.Manifest:
+        manifest = parse_manifest(basedir, configfile, chunk_size, digest_mode, dir_format,
+                                  compress)
         with MultiBar(f"push {repository}@{version}", self.concurrency, quiet=quiet) as mb:
             descs: List[types.Descriptor] = [manifest.config] + list(manifest.blobs)
             for desc in descs:
@@ -126,6 +164,8 @@ class Pusher:
             return os.path.join(basedir, MODELX_CACHE_DIR, desc.name + ".tar.gz")
         if desc.media_type == types.MEDIA_TYPE_MODEL_DIRECTORY_TAR:
             return os.path.join(basedir, MODELX_CACHE_DIR, desc.name + ".tar")
+        if desc.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
+            return os.path.join(basedir, MODELX_CACHE_DIR, desc.name + ".zst")
         return os.path.join(basedir, desc.name)
 
     def push_blob(self, repository: str, desc: types.Descriptor, src_path: str,

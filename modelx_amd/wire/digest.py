@@ -149,6 +149,20 @@ def root_from_leaf_bytes(leaves: bytes, chunk_size: int, total_len: int) -> str:
     return f"{chunked_algo_name(chunk_size)}:{chunked_root(parts, chunk_size, total_len)}"
 
 
+def verify_bytes(data: bytes, digest: str) -> bool:
+    """Verify bytes against a digest string of either algorithm."""
+    try:
+        algo, _ = parse(digest)
+    except ValueError:
+        return False
+    cs = algo_chunk_size(algo)
+    if cs:
+        return chunked_digest(data, cs) == digest
+    if algo == "sha256":
+        return sha256_digest(data) == digest
+    return False
+
+
 class StreamingDigester:
     """Incremental digest over a byte stream, computing BOTH the canonical
     sha256 and the chunked digest in one pass (used by the CPU push path;

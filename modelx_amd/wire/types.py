@@ -46,6 +46,17 @@ MEDIA_TYPE_MODEL_LEAVES = "application/vnd.modelx.amd.leaves.v1"
 # (optional alternative to ...directory.v1.tar+gz; reference clients treat it
 # as an opaque file)
 MEDIA_TYPE_MODEL_DIRECTORY_TAR = "application/vnd.modelx.amd.directory.v1.tar"
+# zstd-compressed file blob: standard multi-frame zstd stream (`zstd -d`
+# decodes it) + zstd-seekable-format seek table, one independent frame per
+# 128 KiB of raw data so the CDNA4 kernels (core/hip/zstd.hip) decode all
+# frames in parallel. The descriptor digest covers the COMPRESSED bytes
+# (content addressing = stored bytes); the annotations below carry the
+# uncompressed identity for post-decompress GPU verification.
+MEDIA_TYPE_MODEL_FILE_ZSTD = "application/vnd.modelx.amd.file.v1+zstd"
+#   chunked digest of the UNCOMPRESSED content (same scheme as chunk-digest)
+ANNOTATION_RAW_DIGEST = "modelx.amd/raw-digest"
+#   uncompressed size, bytes (decimal string)
+ANNOTATION_RAW_SIZE = "modelx.amd/raw-size"
 
 GO_ZERO_TIME = "0001-01-01T00:00:00Z"
 
