@@ -62,7 +62,7 @@ class GpuClient:
 
     def __init__(self, registry: str, authorization: str = "", device: int = 0,
                  num_slots: int = DEFAULT_NUM_SLOTS, slot_bytes: int = DEFAULT_SLOT_BYTES,
-                 num_conns: int = DEFAULT_NUM_CONNS):
+                 num_conns: int = DEFAULT_NUM_CONNS, dedup: bool = False):
         core = _core()
         if not core.hip_available():
             raise er.ModelxError(er.ErrCode.INTERNAL, "no HIP device visible")
@@ -73,6 +73,13 @@ class GpuClient:
         self.engine = core.GpuEngine(device=device, num_slots=num_slots,
                                      slot_bytes=slot_bytes, num_streams=4)
         self.last_stats: List[dict] = []
+        # cross-blob chunk dedup (SURVEY.md §2.2 chunk_verify_dedup): leaf
+        # digest -> (tensor, offset, length) of an HBM-resident chunk. When a
+        # pull's expected leaves hit the index, the chunk is gathered D2D at
+        # HBM bandwidth (~8 TB/s) instead of re-fetched from S3. Opt-in: the
+        # index holds tensor references (keeps them alive).
+        self.dedup = dedup
+        self._chunk_index: Dict[bytes, Tuple[object, int, int]] = {}
 
     # ----------------------------------------------------------- helpers --
 
@@ -145,6 +152,47 @@ class GpuClient:
             else:
                 ranges.append((off, ln))
         return ranges
+
+    # ------------------------------------------------------ chunk dedup --
+
+    def register_chunks(self, tensor, leaves: bytes, chunk_size: int) -> None:
+        """Add a blob's chunks to the dedup index (leaf digest -> location)."""
+        if not self.dedup:
+            return
+        size = tensor.numel() * tensor.element_size()
+        n = len(leaves) // 32
+        for i in range(n):
+            off = i * chunk_size
+            ln = min(chunk_size, size - off)
+            if ln <= 0:
+                break
+            self._chunk_index.setdefault(leaves[i * 32 : (i + 1) * 32], (tensor, off, ln))
+
+    def clear_chunk_index(self) -> None:
+        self._chunk_index.clear()
+
+    def _dedup_plan(self, expect: bytes, chunk_size: int, size: int,
+                    dst_ptr: int) -> Tuple[List[Tuple[int, int, int]], List[Tuple[int, int]]]:
+        """Split a pull into (D2D gather segs for resident chunks, byte
+        ranges to fetch). Gather segs are absolute (src_addr, dst_addr, len)
+        for the generic copy kernel."""
+        segs: List[Tuple[int, int, int]] = []
+        missing: List[Tuple[int, int]] = []
+        n = len(expect) // 32
+        for i in range(n):
+            off = i * chunk_size
+            ln = min(chunk_size, size - off)
+            if ln <= 0:
+                break
+            hit = self._chunk_index.get(expect[i * 32 : (i + 1) * 32])
+            if hit is not None and hit[2] == ln:
+                t, src_off, _ = hit
+                segs.append((t.data_ptr() + src_off, dst_ptr + off, ln))
+            elif missing and missing[-1][0] + missing[-1][1] == off:
+                missing[-1] = (missing[-1][0], missing[-1][1] + ln)
+            else:
+                missing.append((off, ln))
+        return segs, missing
 
     def _fetch_ranges(self, url: str, headers: Dict[str, str], ptr: int,
                       ranges: List[Tuple[int, int]]) -> int:
@@ -224,6 +272,26 @@ class GpuClient:
                 return tensor
             # no leaves sidecar → fall through to a full pull
 
+        if self.dedup and self._chunk_index:
+            # cross-blob dedup: gather HBM-resident chunks D2D, fetch only
+            # the rest (SURVEY.md §2.2 chunk_verify_dedup; the reference
+            # dedups at whole-blob granularity only, push.go:169-177)
+            expect = self._expected_leaves(repository, desc)
+            if expect is not None:
+                segs, missing = self._dedup_plan(expect, cs, desc.size, tensor.data_ptr())
+                if segs:
+                    t0 = time.monotonic()
+                    self.engine.tar_scatter(0, segs)
+                    fetched = self._fetch_ranges(url, headers, tensor.data_ptr(), missing)
+                    self.last_stats.append({
+                        "phase": "pull-dedup", "bytes": fetched,
+                        "dedup_bytes": sum(s[2] for s in segs),
+                        "seconds": time.monotonic() - t0})
+                    if verify:
+                        self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+                    self.register_chunks(tensor, expect, cs)
+                    return tensor
+
         # streaming-hash path: chunks are hashed on each slot's stream right
         # behind its H2D copy, so verification overlaps the transfer
         expect_digest = ""
@@ -245,6 +313,7 @@ class GpuClient:
             if got == expect_digest:
                 self.last_stats.append({"phase": "pull-verify", "bytes": desc.size,
                                         "seconds": time.monotonic() - t0 - stats["seconds"]})
+                self.register_chunks(tensor, leaves, cs)
                 return tensor
             # fall through to the refetch path below
         else:
@@ -463,6 +532,8 @@ class GpuClient:
                              types.ANNOTATION_LEAVES_BLOB: leaves_digest,
                              **extra_notes})
             self.push_blob_from_device(repository, desc, push_ptr, part_bytes=part_bytes)
+            if comp_keep is None:
+                self.register_chunks(t, leaves, chunk_size)
             del comp_keep
             manifest.blobs.append(desc)
             # leaves sidecar: 32 B per chunk, enables chunk-level

@@ -264,3 +264,38 @@ class TestZstdGpuClient:
         finally:
             mdx.stop()
             s3d.stop()
+
+
+class TestChunkDedup:
+    def test_second_pull_dedups(self, tmp_path):
+        """Pulling content whose chunks are already HBM-resident must gather
+        D2D instead of re-fetching (config-5 dedup semantics)."""
+        from util_servers import start_modelxd_s3, start_s3d
+
+        from modelx_amd.client.gpu import GpuClient
+
+        s3d = start_s3d(str(tmp_path / "s3"))
+        mdx = start_modelxd_s3(s3d.url, redirect=True)
+        try:
+            g = GpuClient(mdx.url, device=0, num_slots=4, slot_bytes=8 << 20, dedup=True)
+            base = torch.randint(0, 256, (4 << 20,), dtype=torch.uint8, device="cuda")
+            g.push_from_gpu("gpu/dedup", "v1", {"a.bin": base})
+            # v2 = same chunks under another name + a fresh tail
+            tail = torch.randint(0, 256, (256 << 10,), dtype=torch.uint8, device="cuda")
+            v2 = torch.cat([base, tail])
+            g.push_from_gpu("gpu/dedup", "v2", {"b.bin": v2})
+            g.clear_chunk_index()
+
+            first = g.pull_to_gpu("gpu/dedup", "v1")
+            assert torch.equal(first["a.bin"], base)
+            stats_before = len(g.last_stats)
+            second = g.pull_to_gpu("gpu/dedup", "v2")
+            assert torch.equal(second["b.bin"], v2)
+            dd = [s for s in g.last_stats[stats_before:] if s.get("phase") == "pull-dedup"]
+            assert dd, "dedup path not taken"
+            # all of v1's chunks must have been gathered, only the tail fetched
+            assert dd[0]["dedup_bytes"] >= base.numel()
+            assert dd[0]["bytes"] <= tail.numel() + (128 << 10)
+        finally:
+            mdx.stop()
+            s3d.stop()
