@@ -12,7 +12,28 @@ CORE_SRC = core/src/json.cpp core/src/http.cpp core/src/wire.cpp \
            core/src/store_local.cpp core/src/store_fs.cpp core/src/s3.cpp \
            core/src/sigv4.cpp core/src/auth.cpp core/src/registry.cpp
 
-.PHONY: all clean servers ext test stamp
+.PHONY: all clean servers ext test stamp servers-asan servers-tsan
+
+# sanitizer builds of the C++ servers (SURVEY.md §5 race detection):
+# tests/test_sanitizers.py runs the integration flow against these.
+servers-asan: $(BIN)/modelxd-asan $(BIN)/modelx-s3d-asan
+servers-tsan: $(BIN)/modelxd-tsan $(BIN)/modelx-s3d-tsan
+
+$(BIN)/modelxd-asan: $(CORE_SRC) core/src/modelxd_main.cpp core/include/modelx/*.hpp
+	@mkdir -p $(BIN)
+	$(CXX) $(CXXFLAGS) -g -fsanitize=address $(INCLUDES) $(CORE_SRC) core/src/modelxd_main.cpp -o $@ $(LIBS)
+
+$(BIN)/modelx-s3d-asan: $(CORE_SRC) core/src/s3d.cpp core/include/modelx/*.hpp
+	@mkdir -p $(BIN)
+	$(CXX) $(CXXFLAGS) -g -fsanitize=address $(INCLUDES) $(CORE_SRC) core/src/s3d.cpp -o $@ $(LIBS)
+
+$(BIN)/modelxd-tsan: $(CORE_SRC) core/src/modelxd_main.cpp core/include/modelx/*.hpp
+	@mkdir -p $(BIN)
+	$(CXX) $(CXXFLAGS) -g -fsanitize=thread $(INCLUDES) $(CORE_SRC) core/src/modelxd_main.cpp -o $@ $(LIBS)
+
+$(BIN)/modelx-s3d-tsan: $(CORE_SRC) core/src/s3d.cpp core/include/modelx/*.hpp
+	@mkdir -p $(BIN)
+	$(CXX) $(CXXFLAGS) -g -fsanitize=thread $(INCLUDES) $(CORE_SRC) core/src/s3d.cpp -o $@ $(LIBS)
 
 all: servers ext
 

@@ -27,7 +27,10 @@ struct MxzFrameC {
 
 // ----------------------------------------------------------- decompress ----
 
-__global__ __launch_bounds__(64) void zstd_decompress_frames_kernel(
+// 256 threads (4 waves): every wave redundantly runs the scalar decode so
+// barrier counts line up, and the lane-strided copies go 4x wider — the
+// win on raw-literal frames (our encoder's output), which are copy-bound.
+__global__ __launch_bounds__(256) void zstd_decompress_frames_kernel(
     const u8* __restrict__ src, const MxzFrameC* __restrict__ frames, u32 nframes,
     u8* __restrict__ dst, u8* __restrict__ lit_scratch, i64* __restrict__ rc) {
   __shared__ DecCtx ctx;
@@ -44,7 +47,7 @@ extern "C" hipError_t modelx_zstd_decompress_frames(const void* src, const void*
                                                     void* lit_scratch, int64_t* rc_dev,
                                                     hipStream_t stream) {
   if (nframes == 0) return hipSuccess;
-  hipLaunchKernelGGL(zstd_decompress_frames_kernel, dim3(nframes), dim3(64), 0, stream,
+  hipLaunchKernelGGL(zstd_decompress_frames_kernel, dim3(nframes), dim3(256), 0, stream,
                      static_cast<const u8*>(src), static_cast<const MxzFrameC*>(frames_dev),
                      nframes, static_cast<u8*>(dst), static_cast<u8*>(lit_scratch), rc_dev);
   return hipGetLastError();

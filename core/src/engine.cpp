@@ -569,7 +569,7 @@ class GpuEngine {
     uint64_t nframes = size ? (size + frame_raw - 1) / frame_raw : 1;
     uint64_t stride = (uint64_t)frame_raw + 64 + 3 * ((frame_raw + kBlockMax - 1) / kBlockMax);
     uint32_t max_seqs = kBlockMax / 4 + 1;
-    uint64_t batch = std::min<uint64_t>(nframes, 1024);
+    uint64_t batch = std::min<uint64_t>(nframes, 4096);
     std::vector<zstdhost::SeekEntry> entries(nframes);
     py::gil_scoped_release release;
     void* dscratch = nullptr;

@@ -1,0 +1,237 @@
+"""BASELINE config-4 and config-5 shaped benchmarks (GPU, via gpurun or
+torchrun for multi-rank).
+
+  config4: Llama-3-70B-shaped sharded index — 30 shards, 160 GiB total at
+           --scale 1. Setup pushes the shards; the timed region is the
+           sharded pull (each rank fetches its ShardPlan 1/N, N× S3
+           concurrency) with on-GPU digest verify, plus optional
+           --replicate RCCL broadcast so every rank holds the full set.
+  config5: 2 TiB-shaped mixed index — many small + few huge blobs at
+           --scale 1, zstd-compressed content with cross-blob duplicate
+           chunks. Timed region = pull with GPU zstd decompress + chunk
+           dedup; metric counts LOGICAL (decompressed) bytes landed.
+
+  python tools/bench_shapes.py config4 --scale 0.02 --steps 2
+  torchrun --nproc-per-node 8 tools/bench_shapes.py config4 --scale 1
+
+Single-GPU runs measure the per-GPU data plane (world=1 shard = whole
+index); the driver's 8-GPU round-end bench covers scaling of the headline
+metric; this tool documents the config-4/5 shapes (profiles/bench_shapes.md).
+"""
+import argparse
+import json
+import os
+import shutil
+import sys
+import time
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "tests"))
+
+
+def start_stack(rank: int, store_root: str):
+    from util_servers import MODELXD, S3D, ServerProc, wait_http
+
+    base = int(os.environ.get("MASTER_PORT", "29500"))
+    s3_port = base + 2371 + 2 * rank
+    mdx_port = base + 2372 + 2 * rank
+    shutil.rmtree(store_root, ignore_errors=True)
+    os.makedirs(os.path.join(store_root, "modelx"), exist_ok=True)
+    procs = [ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root", store_root,
+                         "--access-key", "modelx", "--secret-key", "modelx123"], s3_port)]
+    wait_http(s3_port)
+    procs.append(ServerProc(
+        [MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
+         f"http://127.0.0.1:{s3_port}", "--s3-bucket", "modelx", "--s3-access-key", "modelx",
+         "--s3-secret-key", "modelx123", "--enable-redirect"], mdx_port))
+    wait_http(mdx_port)
+    return procs, f"http://127.0.0.1:{mdx_port}"
+
+
+def config4(args, g, dist, rank, world, device):
+    """Sharded Llama-70B-shape: 30 shards, 160 GiB at scale 1."""
+    import torch
+
+    from modelx_amd.client.fanout import ShardPlan, fanout_pull_sharded
+
+    from modelx_amd.wire import types as t
+
+    total = int(160 * args.scale * (1 << 30))
+    nshards = 30
+    shard = max(total // nshards, 16 << 20)
+    names = [f"model-{i:05d}-of-{nshards:05d}.safetensors" for i in range(nshards)]
+    plan = ShardPlan.build([type("D", (), {"name": n, "size": shard})() for n in names], world)
+    # setup (untimed): each rank pushes the shards it owns, one version per
+    # shard; rank 0 then publishes the combined "all" manifest
+    src = torch.empty(shard, dtype=torch.uint8, device=f"cuda:{device}")
+    for i, n in enumerate(names):
+        if plan.owners[n] != rank:
+            continue
+        torch.manual_seed(i)
+        src.random_(0, 256)
+        g.push_from_gpu("bench/llama70b", n, {n: src})
+    del src
+    torch.cuda.empty_cache()
+    if dist:
+        dist.barrier()
+    if rank == 0:
+        m = t.Manifest(media_type=t.MEDIA_TYPE_MODEL_MANIFEST_JSON)
+        seen = {}
+        for n in names:
+            mm = g.remote.get_manifest("bench/llama70b", n)
+            m.config = mm.config
+            for b in mm.blobs:
+                seen[b.name] = b
+        m.blobs = t.sort_descriptors_by_name(list(seen.values()))
+        g.remote.put_manifest("bench/llama70b", "all", m)
+    if dist:
+        dist.barrier()
+
+    def sync():
+        torch.cuda.synchronize(device)
+        if dist:
+            dist.barrier()
+        torch.cuda.synchronize(device)
+
+    landed = 0
+    sync()
+    t0 = time.monotonic()
+    for _ in range(args.steps):
+        out = fanout_pull_sharded(dist, g, "bench/llama70b", "all", device,
+                                  replicate=args.replicate) if dist else None
+        if not dist:
+            m = g.remote.get_manifest("bench/llama70b", "all")
+            out = {}
+            for d in m.blobs:
+                if d.size and d.media_type == t.MEDIA_TYPE_MODEL_FILE:
+                    out[d.name] = g.pull_blob_to_device("bench/llama70b", d)
+        landed += sum(v.numel() for v in out.values())
+        del out
+        torch.cuda.empty_cache()
+    sync()
+    dt = time.monotonic() - t0
+    return {"metric": "config4 llama70b sharded pull GiB/s",
+            "value": round(landed * (world if args.replicate else 1) / dt / (1 << 30), 3),
+            "per_rank_landed_gib": round(landed / (1 << 30), 2),
+            "shards": nshards, "shard_gib": round(shard / (1 << 30), 3)}
+
+
+def config5(args, g, dist, rank, world, device):
+    """Mixed 2 TiB-shape index: many small + few huge, zstd + dedup."""
+    import torch
+
+    # at scale 1 per node: 1792 small x 64 MiB + 16 huge x 120 GiB ~= 2 TiB
+    # per rank (of 8): 224 small + 2 huge
+    small_n = max(int(224 * args.scale * 8 / max(world, 1)), 2)
+    small_sz = 64 << 20
+    huge_n = 2
+    huge_sz = max(int(120 * args.scale * (1 << 30)), 64 << 20)
+    page = torch.randint(0, 256, (1 << 20,), dtype=torch.uint8, device=f"cuda:{device}")
+
+    def make(nbytes: int, seed: int) -> torch.Tensor:
+        # tiled shared pages (dedup + compressible) + unique head
+        reps = nbytes // page.numel() + 1
+        tns = page.repeat(reps)[:nbytes].contiguous()
+        torch.manual_seed(seed)
+        head = torch.randint(0, 256, (min(nbytes, 4 << 20),), dtype=torch.uint8,
+                             device=f"cuda:{device}")
+        tns[: head.numel()] = head
+        return tns
+
+    # small blobs ride the zstd path (GPU decompress); huge blobs stay raw so
+    # chunk dedup applies to the stored bytes (shared page tiles)
+    repo = f"bench/mixed-r{rank}"
+    for i in range(small_n):
+        t_small = make(small_sz, 1000 + i)
+        g.push_from_gpu(repo, f"s{i}", {"blob.bin": t_small}, compress="zstd")
+        del t_small
+    for i in range(huge_n):
+        t_huge = make(huge_sz, 2000 + i)
+        g.push_from_gpu(repo, f"h{i}", {"blob.bin": t_huge})
+        del t_huge
+    torch.cuda.empty_cache()
+
+    def sync():
+        torch.cuda.synchronize(device)
+        if dist:
+            dist.barrier()
+        torch.cuda.synchronize(device)
+
+    g.dedup = True
+    g.clear_chunk_index()
+    sync()
+    t0 = time.monotonic()
+    landed = 0
+    for _ in range(args.steps):
+        g.clear_chunk_index()
+        for i in range(small_n):
+            out = g.pull_to_gpu(repo, f"s{i}")
+            landed += sum(v.numel() for v in out.values())
+            del out
+        for i in range(huge_n):
+            out = g.pull_to_gpu(repo, f"h{i}")
+            landed += sum(v.numel() for v in out.values())
+            del out
+        torch.cuda.empty_cache()
+    sync()
+    dt = time.monotonic() - t0
+    import collections
+
+    agg = collections.defaultdict(lambda: [0.0, 0])
+    for s in g.last_stats:
+        agg[s.get("phase", "?")][0] += s.get("seconds", 0.0)
+        agg[s.get("phase", "?")][1] += s.get("bytes", 0)
+    stages = {k: {"s": round(v[0], 2), "gib": round(v[1] / (1 << 30), 2)}
+              for k, v in sorted(agg.items())}
+    return {"metric": "config5 mixed-index logical pull GiB/s (zstd+dedup)",
+            "value": round(landed * world / dt / (1 << 30), 3),
+            "per_rank_logical_gib": round(landed / (1 << 30), 2),
+            "small_blobs": small_n, "huge_gib": round(huge_sz / (1 << 30), 2),
+            "stages": stages}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("config", choices=["config4", "config5"])
+    ap.add_argument("--scale", type=float, default=0.02,
+                    help="1.0 = full BASELINE shape (160 GiB / 2 TiB per node)")
+    ap.add_argument("--steps", type=int, default=1)
+    ap.add_argument("--replicate", action="store_true")
+    ap.add_argument("--conns", type=int, default=8)
+    args = ap.parse_args()
+
+    import torch
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    device = int(os.environ.get("LOCAL_RANK", str(rank)))
+    torch.cuda.set_device(device)
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+
+        tdist.init_process_group(backend="nccl")
+        dist = tdist
+
+    store = ("/dev/shm" if os.path.isdir("/dev/shm") else "/tmp") + f"/modelx-shape-r{rank}"
+    procs, url = start_stack(rank, store)
+    from modelx_amd.client.gpu import GpuClient
+
+    g = GpuClient(url, device=device, num_conns=args.conns)
+    try:
+        fn = config4 if args.config == "config4" else config5
+        result = fn(args, g, dist, rank, world, device)
+        result.update({"n_gpus": world, "scale": args.scale, "steps": args.steps})
+        if rank == 0:
+            print(json.dumps(result), flush=True)
+    finally:
+        for p in procs:
+            p.stop()
+        shutil.rmtree(store, ignore_errors=True)
+        if dist:
+            dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -8,10 +8,11 @@ roundtrip. rocprofv3 over this script gives per-kernel evidence
 (zstd_compress_frames_kernel / zstd_decompress_frames_kernel).
 """
 import argparse
+import os
 import sys
 import time
 
-sys.path.insert(0, ".")
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
