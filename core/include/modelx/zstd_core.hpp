@@ -231,9 +231,14 @@ struct FseTable {
 // Table-construction temporaries. On the GPU these live in LDS (inside
 // DecCtx) so the redundant-wavefront execution doesn't spill 64 private
 // copies to scratch memory; on the CPU they're just part of the context.
+// INVARIANT (multi-wave workgroups): every write to these shared tables
+// must be a deterministic same-value write — all threads compute the same
+// value for the same cell — because the 4 waves of a 256-thread decompress
+// workgroup are NOT in lockstep between barriers. Read-modify-write state
+// (occurrence counters, repeat offsets) therefore lives in per-thread
+// locals, never in this struct.
 struct BuildScratch {
   u16 spread[kMaxFseSize];  // cell -> symbol during spread
-  u16 counter[256];         // per-symbol occurrence counters
   u8 weights[256];          // huffman weights
   i16 counts[256];          // normalized counts from fse_read_ncount
   FseTable wtab;            // huffman-weight FSE table
@@ -262,8 +267,10 @@ MX_HD static inline int fse_build_dtable(FseTable* t, const i16* counts, u32 nsy
     }
   }
   if (pos != 0) return MXZ_ERR_FSE;
-  // per-symbol occurrence counters start at the normalized count
-  u16* counter = bs->counter;
+  // per-symbol occurrence counters start at the normalized count.
+  // Thread-LOCAL: counter[s]++ is a read-modify-write and must not be
+  // shared across the workgroup's waves (see BuildScratch invariant).
+  u16 counter[256];
   for (u32 s = 0; s < nsym; s++) counter[s] = (u16)(counts[s] == -1 ? 1 : (counts[s] < 0 ? 0 : counts[s]));
   for (u32 c = 0; c < size; c++) {
     u32 s = pos_syms[c];
@@ -534,16 +541,12 @@ struct DecCtx {
   FseTable ll, of, ml;  // persist across blocks (repeat mode)
   BuildScratch bs;      // table-construction temporaries
   bool ll_valid, of_valid, ml_valid;
-  u32 rep[3];  // repeat offsets
   u8* lit_scratch;      // >= kBlockMax bytes
 };
 
 MX_HD static inline void dec_ctx_init(DecCtx* c, u8* lit_scratch) {
   c->huf.valid = false;
   c->ll_valid = c->of_valid = c->ml_valid = false;
-  c->rep[0] = 1;
-  c->rep[1] = 4;
-  c->rep[2] = 8;
   c->lit_scratch = lit_scratch;
 }
 
@@ -592,8 +595,8 @@ MX_HD static inline int seq_table_load(FseTable* t, bool* valid, u32 mode, u32 k
 
 // Decode one compressed block into dst (history = bytes already written to
 // the frame buffer before dst). Returns regenerated size or <0.
-MX_HD static inline i64 decode_block(DecCtx* c, const u8* src, u64 srclen, u8* dst, u64 dstcap,
-                                     u64 history) {
+MX_HD static inline i64 decode_block(DecCtx* c, u32* rep, const u8* src, u64 srclen, u8* dst,
+                                     u64 dstcap, u64 history) {
   // ---- literals section ----
   if (srclen < 1) return MXZ_ERR_LITERALS;
   u32 b0 = src[0];
@@ -766,19 +769,19 @@ MX_HD static inline i64 decode_block(DecCtx* c, const u8* src, u64 srclen, u8* d
     if (ofValue <= 3) {
       u32 idx = (u32)ofValue - 1 + (ll == 0 ? 1 : 0);
       if (idx == 0) {
-        offset = c->rep[0];
+        offset = rep[0];
       } else {
-        offset = idx < 3 ? c->rep[idx] : (u64)c->rep[0] - 1;
+        offset = idx < 3 ? rep[idx] : (u64)rep[0] - 1;
         if (offset == 0) return MXZ_ERR_OFFSET;
-        if (idx > 1) c->rep[2] = c->rep[1];
-        c->rep[1] = c->rep[0];
-        c->rep[0] = (u32)offset;
+        if (idx > 1) rep[2] = rep[1];
+        rep[1] = rep[0];
+        rep[0] = (u32)offset;
       }
     } else {
       offset = ofValue - 3;
-      c->rep[2] = c->rep[1];
-      c->rep[1] = c->rep[0];
-      c->rep[0] = (u32)offset;
+      rep[2] = rep[1];
+      rep[1] = rep[0];
+      rep[0] = (u32)offset;
     }
     // copy literals
     if (lit_used + ll > lit_regen) return MXZ_ERR_SEQUENCES;
@@ -868,6 +871,7 @@ MX_HD static inline i64 decode_frame(const u8* src, u64 srclen, u8* dst, u64 dst
   if (have_fcs && fcs > dstcap) return MXZ_ERR_DST_SMALL;
 
   dec_ctx_init(ctx, ctx->lit_scratch);
+  u32 rep[3] = {1, 4, 8};  // thread-local: RMW state (see BuildScratch note)
   u64 out = 0;
   while (true) {
     if (pos + 3 > srclen) return MXZ_ERR_SRC_TRUNC;
@@ -891,7 +895,7 @@ MX_HD static inline i64 decode_frame(const u8* src, u64 srclen, u8* dst, u64 dst
     } else if (btype == 2) {
       if (bsize > kBlockMax + 32) return MXZ_ERR_BLOCK;
       if (pos + bsize > srclen) return MXZ_ERR_SRC_TRUNC;
-      i64 n = decode_block(ctx, src + pos, bsize, dst + out, dstcap - out, out);
+      i64 n = decode_block(ctx, rep, src + pos, bsize, dst + out, dstcap - out, out);
       if (n < 0) return n;
       out += (u64)n;
       pos += bsize;
