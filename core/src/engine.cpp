@@ -169,15 +169,15 @@ class GpuEngine {
     HIP_CHECK(hipSetDevice(device_));
     double t0 = now_s();
 
-    // build the range queue (slot-sized)
-    {
-      std::lock_guard<std::mutex> lk(mu_);
-      ranges_.clear();
-      for (uint64_t off = 0; off < size; off += slot_bytes_)
-        ranges_.push_back({off, std::min<uint64_t>(slot_bytes_, size - off)});
-      next_range_ = 0;
-      error_.clear();
-    }
+    // per-call pull state: pull_impl is reentrant — concurrent pulls (e.g.
+    // many small blobs of one index from a Python thread pool) share the
+    // pinned-slot pool and streams but own their range queue and error.
+    std::vector<Range> ranges;
+    for (uint64_t off = 0; off < size; off += slot_bytes_)
+      ranges.push_back({off, std::min<uint64_t>(slot_bytes_, size - off)});
+    std::atomic<size_t> next_range{0};
+    std::mutex err_mu;
+    std::string error;
     std::atomic<uint64_t> net_bytes{0};
     std::atomic<long> net_ns{0};
 
@@ -216,9 +216,13 @@ class GpuEngine {
         while (true) {
           Range r;
           {
-            std::lock_guard<std::mutex> lk(mu_);
-            if (!error_.empty() || next_range_ >= ranges_.size()) break;
-            r = ranges_[next_range_++];
+            size_t i = next_range.fetch_add(1);
+            if (i >= ranges.size()) break;
+            {
+              std::lock_guard<std::mutex> lk(err_mu);
+              if (!error.empty()) break;
+            }
+            r = ranges[i];
           }
           Slot* slot = acquire_slot();
           bool ok = false;
@@ -232,11 +236,10 @@ class GpuEngine {
           }
           if (!ok) {
             {
-              std::lock_guard<std::mutex> lk(mu_);
-              if (error_.empty()) error_ = "range fetch failed @" + std::to_string(r.offset);
-              free_.push(slot);
+              std::lock_guard<std::mutex> lk(err_mu);
+              if (error.empty()) error = "range fetch failed @" + std::to_string(r.offset);
             }
-            cv_free_.notify_one();
+            release_slot(slot);
             break;
           }
           hipStream_t st = streams_[stream_rr.fetch_add(1) % streams_.size()];
@@ -254,16 +257,18 @@ class GpuEngine {
           }
           if (e == hipSuccess) e = hipEventRecord(slot->event, st);
           if (e != hipSuccess) {
-            std::lock_guard<std::mutex> lk(mu_);
-            if (error_.empty()) error_ = std::string("hip: ") + hipGetErrorString(e);
-            free_.push(slot);
+            {
+              std::lock_guard<std::mutex> lk(err_mu);
+              if (error.empty()) error = std::string("hip: ") + hipGetErrorString(e);
+            }
+            release_slot(slot);
             break;
           }
           {
             std::lock_guard<std::mutex> lk(mu_);
             pending_.push_back(slot);
           }
-          cv_pending_.notify_one();
+          cv_pending_.notify_all();
         }
       });
     }
@@ -280,8 +285,8 @@ class GpuEngine {
     }
     double t1 = now_s();
     {
-      std::lock_guard<std::mutex> lk(mu_);
-      if (!error_.empty()) throw std::runtime_error("pull_to_device: " + error_);
+      std::lock_guard<std::mutex> lk(err_mu);
+      if (!error.empty()) throw std::runtime_error("pull_to_device: " + error);
     }
     py::gil_scoped_acquire acquire;
     py::dict stats;
@@ -746,9 +751,6 @@ class GpuEngine {
   hipStream_t hash_stream_ = nullptr;
   std::mutex mu_;
   std::condition_variable cv_free_, cv_pending_;
-  std::vector<Range> ranges_;
-  size_t next_range_ = 0;
-  std::string error_;
 };
 
 bool hip_available() {

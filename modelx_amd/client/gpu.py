@@ -359,6 +359,21 @@ class GpuClient:
                 out[desc.name] = self.pull_blob_to_device(repository, desc, verify=verify)
         return out
 
+    def pull_many(self, repository: str, versions, parallel: int = 6,
+                  verify: bool = True) -> Dict[str, Dict[str, "torch.Tensor"]]:
+        """Pull several versions concurrently. The native engine's pull path
+        is reentrant (per-call range state, shared pinned-slot pool), so
+        many small blobs overlap their HTTP round-trips — the limiter for
+        mixed indexes (BASELINE config 5) is per-blob latency, not
+        bandwidth."""
+        from concurrent.futures import ThreadPoolExecutor
+
+        def one(v):
+            return v, self.pull_to_gpu(repository, v, verify=verify)
+
+        with ThreadPoolExecutor(max_workers=parallel) as ex:
+            return dict(ex.map(one, versions))
+
     # -------------------------------------------------- directory blobs --
 
     def pull_dir_to_gpu(self, repository: str, desc: types.Descriptor

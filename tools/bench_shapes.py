@@ -165,10 +165,9 @@ def config5(args, g, dist, rank, world, device):
     landed = 0
     for _ in range(args.steps):
         g.clear_chunk_index()
-        for i in range(small_n):
-            out = g.pull_to_gpu(repo, f"s{i}")
-            landed += sum(v.numel() for v in out.values())
-            del out
+        outs = g.pull_many(repo, [f"s{i}" for i in range(small_n)], parallel=args.parallel)
+        landed += sum(v.numel() for vs in outs.values() for v in vs.values())
+        del outs
         for i in range(huge_n):
             out = g.pull_to_gpu(repo, f"h{i}")
             landed += sum(v.numel() for v in out.values())
@@ -199,6 +198,7 @@ def main():
     ap.add_argument("--steps", type=int, default=1)
     ap.add_argument("--replicate", action="store_true")
     ap.add_argument("--conns", type=int, default=8)
+    ap.add_argument("--parallel", type=int, default=8, help="concurrent blob pulls (small blobs)")
     args = ap.parse_args()
 
     import torch
