@@ -120,6 +120,8 @@ class GpuEngine {
   }
 
   ~GpuEngine() {
+    for (auto& p : zs_ptr_)
+      if (p) hipFree(p);
     for (auto& st : streams_) hipStreamDestroy(st);
     hipStreamDestroy(hash_stream_);
     for (auto& s : slots_) {
@@ -137,7 +139,26 @@ class GpuEngine {
                           const std::map<std::string, std::string>& headers, uint64_t size,
                           uintptr_t dst_ptr, int num_conns, uint64_t base_offset = 0) {
     std::string leaves;
-    return pull_impl(url, headers, size, dst_ptr, num_conns, base_offset, 0, &leaves);
+    return pull_impl(url, headers, size, dst_ptr, num_conns, base_offset, 0, &leaves, nullptr);
+  }
+
+  // Fetch an explicit set of (offset, length) byte ranges of one object in
+  // a single engine pass (shared connection workers + pinned ring) — the
+  // chunk-dedup / chunk-refetch path, where spinning up the full pull
+  // machinery per contiguous range would dominate.
+  py::dict pull_ranges_to_device(const std::string& url,
+                                 const std::map<std::string, std::string>& headers,
+                                 const std::vector<std::pair<uint64_t, uint64_t>>& ranges,
+                                 uintptr_t dst_ptr, int num_conns) {
+    std::string leaves;
+    std::vector<Range> rs;
+    uint64_t total = 0;
+    for (auto& r : ranges) {
+      for (uint64_t off = 0; off < r.second; off += slot_bytes_)
+        rs.push_back({r.first + off, std::min<uint64_t>(slot_bytes_, r.second - off)});
+      total += r.second;
+    }
+    return pull_impl(url, headers, total, dst_ptr, num_conns, 0, 0, &leaves, &rs);
   }
 
   // Same, but ALSO hashes every landed slot with the CDNA4 SHA-256 chunk
@@ -150,13 +171,15 @@ class GpuEngine {
                                   uint64_t size, uintptr_t dst_ptr, int num_conns,
                                   uint64_t chunk_size) {
     std::string leaves;
-    py::dict stats = pull_impl(url, headers, size, dst_ptr, num_conns, 0, chunk_size, &leaves);
+    py::dict stats =
+        pull_impl(url, headers, size, dst_ptr, num_conns, 0, chunk_size, &leaves, nullptr);
     return py::make_tuple(stats, py::bytes(leaves));
   }
 
   py::dict pull_impl(const std::string& url, const std::map<std::string, std::string>& headers,
                      uint64_t size, uintptr_t dst_ptr, int num_conns, uint64_t base_offset,
-                     uint64_t hash_chunk, std::string* leaves_out) {
+                     uint64_t hash_chunk, std::string* leaves_out,
+                     const std::vector<Range>* explicit_ranges) {
     void* dleaves = nullptr;
     uint32_t total_chunks = 0;
     if (hash_chunk) {
@@ -173,8 +196,12 @@ class GpuEngine {
     // many small blobs of one index from a Python thread pool) share the
     // pinned-slot pool and streams but own their range queue and error.
     std::vector<Range> ranges;
-    for (uint64_t off = 0; off < size; off += slot_bytes_)
-      ranges.push_back({off, std::min<uint64_t>(slot_bytes_, size - off)});
+    if (explicit_ranges) {
+      ranges = *explicit_ranges;
+    } else {
+      for (uint64_t off = 0; off < size; off += slot_bytes_)
+        ranges.push_back({off, std::min<uint64_t>(slot_bytes_, size - off)});
+    }
     std::atomic<size_t> next_range{0};
     std::mutex err_mu;
     std::string error;
@@ -566,6 +593,20 @@ class GpuEngine {
   // One workgroup per frame (core/hip/zstd.hip); frames are compressed into
   // padded per-frame scratch in batches, packed with the scatter kernel,
   // and the seek table is appended from the host.
+  // Grow-only device scratch shared by the zstd entry points (guarded by
+  // zstd_mu_): per-call hipMalloc/hipFree of GiB-scale scratch dominated
+  // concurrent decompress (config-5 profile) before this.
+  void* zstd_scratch(size_t idx, size_t need) {
+    if (zs_size_[idx] < need) {
+      if (zs_ptr_[idx]) HIP_CHECK(hipFree(zs_ptr_[idx]));
+      zs_ptr_[idx] = nullptr;
+      zs_size_[idx] = 0;
+      HIP_CHECK(hipMalloc(&zs_ptr_[idx], need));
+      zs_size_[idx] = need;
+    }
+    return zs_ptr_[idx];
+  }
+
   uint64_t zstd_compress_device(uintptr_t src_ptr, uint64_t size, uint32_t frame_raw,
                                 uintptr_t dst_ptr, uint64_t dst_cap) {
     HIP_CHECK(hipSetDevice(device_));
@@ -577,14 +618,11 @@ class GpuEngine {
     uint64_t batch = std::min<uint64_t>(nframes, 4096);
     std::vector<zstdhost::SeekEntry> entries(nframes);
     py::gil_scoped_release release;
-    void* dscratch = nullptr;
-    void* dseqs = nullptr;
-    int64_t* dsizes = nullptr;
-    void* dsegs = nullptr;
-    HIP_CHECK(hipMalloc(&dscratch, batch * stride));
-    HIP_CHECK(hipMalloc(&dseqs, batch * (uint64_t)max_seqs * 12));
-    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dsizes), batch * sizeof(int64_t)));
-    HIP_CHECK(hipMalloc(&dsegs, batch * sizeof(CopySegHost)));
+    std::lock_guard<std::mutex> zlk(zstd_mu_);
+    void* dscratch = zstd_scratch(0, batch * stride);
+    void* dseqs = zstd_scratch(1, batch * (uint64_t)max_seqs * 12);
+    int64_t* dsizes = reinterpret_cast<int64_t*>(zstd_scratch(2, batch * sizeof(int64_t)));
+    void* dsegs = zstd_scratch(3, batch * sizeof(CopySegHost));
     std::vector<int64_t> hsizes(batch);
     std::vector<CopySegHost> hsegs(batch);
     uint64_t out = 0;
@@ -618,10 +656,6 @@ class GpuEngine {
     HIP_CHECK(hipMemcpy(reinterpret_cast<char*>(dst_ptr) + out, table.data(), table.size(),
                         hipMemcpyHostToDevice));
     out += table.size();
-    hipFree(dscratch);
-    hipFree(dseqs);
-    hipFree(dsizes);
-    hipFree(dsegs);
     return out;
   }
 
@@ -666,12 +700,10 @@ class GpuEngine {
     }
     if (d_off > dst_cap) throw std::runtime_error("zstd decompress: dst too small");
     uint64_t batch = std::min<uint64_t>(nframes ? nframes : 1, 8192);
-    void* dframes = nullptr;
-    void* dlit = nullptr;
-    int64_t* drc = nullptr;
-    HIP_CHECK(hipMalloc(&dframes, batch * sizeof(MxzFrameHost)));
-    HIP_CHECK(hipMalloc(&dlit, batch * (uint64_t)kBlockMax));
-    HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&drc), batch * sizeof(int64_t)));
+    std::lock_guard<std::mutex> zlk(zstd_mu_);
+    void* dframes = zstd_scratch(4, batch * sizeof(MxzFrameHost));
+    void* dlit = zstd_scratch(5, batch * (uint64_t)kBlockMax);
+    int64_t* drc = reinterpret_cast<int64_t*>(zstd_scratch(6, batch * sizeof(int64_t)));
     std::vector<int64_t> hrc(batch);
     for (uint64_t first = 0; first < nframes; first += batch) {
       uint32_t n = (uint32_t)std::min<uint64_t>(batch, nframes - first);
@@ -687,9 +719,6 @@ class GpuEngine {
           throw std::runtime_error("zstd decompress kernel failed: frame " +
                                    std::to_string(first + i) + " rc=" + std::to_string(hrc[i]));
     }
-    hipFree(dframes);
-    hipFree(dlit);
-    hipFree(drc);
     return d_off;
   }
 
@@ -751,6 +780,9 @@ class GpuEngine {
   hipStream_t hash_stream_ = nullptr;
   std::mutex mu_;
   std::condition_variable cv_free_, cv_pending_;
+  std::mutex zstd_mu_;
+  void* zs_ptr_[7] = {};
+  size_t zs_size_[7] = {};
 };
 
 bool hip_available() {
@@ -777,6 +809,8 @@ PYBIND11_MODULE(_core, m) {
       .def("pull_to_device", &GpuEngine::pull_to_device, py::arg("url"), py::arg("headers"),
            py::arg("size"), py::arg("dst_ptr"), py::arg("num_conns") = 8,
            py::arg("base_offset") = 0)
+      .def("pull_ranges_to_device", &GpuEngine::pull_ranges_to_device, py::arg("url"),
+           py::arg("headers"), py::arg("ranges"), py::arg("dst_ptr"), py::arg("num_conns") = 8)
       .def("pull_to_device_hashed", &GpuEngine::pull_to_device_hashed, py::arg("url"),
            py::arg("headers"), py::arg("size"), py::arg("dst_ptr"), py::arg("num_conns") = 8,
            py::arg("chunk_size") = (uint64_t)(128 << 10))

@@ -87,38 +87,10 @@ bool write_stream_to(const std::string& path, http::Request& req, std::string* e
   int fd = ::open(tmp.c_str(), O_RDWR | O_CREAT | O_TRUNC | O_CLOEXEC, 0644);
   if (fd < 0) return false;
   int64_t total = 0;
-  // known length: ftruncate + mmap and recv straight into the page cache —
-  // one copy (socket → mapped file) instead of two (socket → buffer →
-  // write). The writable mapping also leaves the pages hot for the GET that
-  // typically follows (the push→pull pattern), instead of faulting them in
-  // again on first read.
-  if (req.content_length > 0 && ::ftruncate(fd, req.content_length) == 0) {
-    void* map = ::mmap(nullptr, static_cast<size_t>(req.content_length),
-                       PROT_READ | PROT_WRITE, MAP_SHARED, fd, 0);
-    if (map != MAP_FAILED) {
-      char* p = static_cast<char*>(map);
-      while (total < req.content_length) {
-        size_t want = static_cast<size_t>(
-            std::min<int64_t>(req.content_length - total, 16 << 20));
-        ssize_t r = req.read_body(p + total, want);
-        if (r <= 0) break;
-        total += r;
-      }
-      ::munmap(map, static_cast<size_t>(req.content_length));
-      ::close(fd);
-      if (total != req.content_length) {
-        ::unlink(tmp.c_str());
-        return false;
-      }
-      if (::rename(tmp.c_str(), path.c_str()) != 0) {
-        ::unlink(tmp.c_str());
-        return false;
-      }
-      if (etag) *etag = "\"s3d-" + std::to_string(total) + "\"";
-      return true;
-    }
-    // mmap failed → fall through to the write() loop
-  }
+  // NOTE: an mmap+recv variant (ftruncate, MAP_SHARED, recv into the
+  // mapping) was A/B-tested on the GPU box and LOST to plain write() on the
+  // tmpfs store: per-page fault handling in the recv copy path costs more
+  // than the buffer bounce it saves (push throughput 7.5 vs 12.5 GiB/s).
   std::vector<char> buf(4 << 20);
   while (true) {
     ssize_t r = req.read_body(buf.data(), buf.size());

@@ -196,11 +196,10 @@ class GpuClient:
 
     def _fetch_ranges(self, url: str, headers: Dict[str, str], ptr: int,
                       ranges: List[Tuple[int, int]]) -> int:
-        total = 0
-        for off, ln in ranges:
-            self.engine.pull_to_device(url, headers, ln, ptr + off, self.num_conns, off)
-            total += ln
-        return total
+        if not ranges:
+            return 0
+        self.engine.pull_ranges_to_device(url, headers, ranges, ptr, self.num_conns)
+        return sum(ln for _, ln in ranges)
 
     def pull_zstd_blob_to_device(self, repository: str, desc: types.Descriptor,
                                  verify: bool = True) -> "torch.Tensor":
