@@ -227,7 +227,7 @@ class GpuEngine {
           std::lock_guard<std::mutex> lk(mu_);
           free_.push(s);
         }
-        cv_free_.notify_one();
+        cv_free_.notify_all();
       }
     });
 
@@ -400,8 +400,18 @@ class GpuEngine {
     // part uploads don't serialize on one stream
     hipStream_t st_a = streams_[push_rr_.fetch_add(1) % streams_.size()];
     hipStream_t st_b = streams_[push_rr_.fetch_add(1) % streams_.size()];
-    Slot* cur = acquire_slot();
-    Slot* nxt = acquire_slot();
+    // both slots in one wait: sequential acquire_slot() would hold-and-wait
+    // and can deadlock when >= num_slots pushes run concurrently
+    Slot* cur = nullptr;
+    Slot* nxt = nullptr;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      cv_free_.wait(lk, [&] { return free_.size() >= 2; });
+      cur = free_.front();
+      free_.pop();
+      nxt = free_.front();
+      free_.pop();
+    }
     uint64_t off = 0;
     uint64_t cur_len = std::min<uint64_t>(slot_bytes_, size);
     HIP_CHECK(hipMemcpyAsync(cur->host, reinterpret_cast<char*>(src_ptr), cur_len,
@@ -750,7 +760,7 @@ class GpuEngine {
       std::lock_guard<std::mutex> lk(mu_);
       free_.push(s);
     }
-    cv_free_.notify_one();
+    cv_free_.notify_all();
   }
 
   // ranged GET into `dst` (pinned); returns false on any protocol error

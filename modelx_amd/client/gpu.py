@@ -344,19 +344,27 @@ class GpuClient:
         return tensor
 
     def pull_to_gpu(self, repository: str, version: str = "",
-                    verify: bool = True) -> Dict[str, "torch.Tensor"]:
+                    verify: bool = True, parallel: int = 1) -> Dict[str, "torch.Tensor"]:
         """Pull every file blob of a manifest into HBM. Directory (tar.gz)
-        blobs are landed as raw archive bytes under their blob name."""
+        blobs are landed as raw archive bytes under their blob name.
+        ``parallel`` > 1 pulls blobs concurrently (reentrant engine; right
+        for many-shard manifests where per-blob latency would stack)."""
         manifest = self.remote.get_manifest(repository, version)
-        out: Dict[str, "torch.Tensor"] = {}
-        for desc in manifest.blobs:
-            if desc.size == 0 or desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
-                continue
+        descs = [d for d in manifest.blobs
+                 if d.size and d.media_type != types.MEDIA_TYPE_MODEL_LEAVES]
+
+        def one(desc):
             if desc.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
-                out[desc.name] = self.pull_zstd_blob_to_device(repository, desc, verify=verify)
-            else:
-                out[desc.name] = self.pull_blob_to_device(repository, desc, verify=verify)
-        return out
+                return desc.name, self.pull_zstd_blob_to_device(repository, desc,
+                                                                verify=verify)
+            return desc.name, self.pull_blob_to_device(repository, desc, verify=verify)
+
+        if parallel > 1 and len(descs) > 1:
+            from concurrent.futures import ThreadPoolExecutor
+
+            with ThreadPoolExecutor(max_workers=parallel) as ex:
+                return dict(ex.map(one, descs))
+        return dict(one(d) for d in descs)
 
     def pull_many(self, repository: str, versions, parallel: int = 6,
                   verify: bool = True) -> Dict[str, Dict[str, "torch.Tensor"]]:

@@ -101,11 +101,7 @@ def config4(args, g, dist, rank, world, device):
         out = fanout_pull_sharded(dist, g, "bench/llama70b", "all", device,
                                   replicate=args.replicate) if dist else None
         if not dist:
-            m = g.remote.get_manifest("bench/llama70b", "all")
-            out = {}
-            for d in m.blobs:
-                if d.size and d.media_type == t.MEDIA_TYPE_MODEL_FILE:
-                    out[d.name] = g.pull_blob_to_device("bench/llama70b", d)
+            out = g.pull_to_gpu("bench/llama70b", "all", parallel=args.parallel)
         landed += sum(v.numel() for v in out.values())
         del out
         torch.cuda.empty_cache()
