@@ -72,6 +72,10 @@ class ResponseWriter {
   void begin(int status, int64_t content_length);
   void write(const char* data, size_t n);
   bool sendfile(int fd, int64_t offset, int64_t count);  // after begin()
+  // Hard-fail a partially-sent response: marks the writer failed and shuts
+  // the socket down so the peer sees EOF mid-body instead of a desynced
+  // stream (next response bytes read as body).
+  void abort_connection();
   bool sent() const { return sent_; }
   bool failed() const { return failed_; }
 

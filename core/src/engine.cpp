@@ -772,12 +772,22 @@ class GpuEngine {
     int status = 0;
     http::Headers rh;
     if (!conn.read_response_head(&status, &rh)) return false;
-    if (status != 206 && status != 200) {
+    // STRICT framing: a ranged GET must answer 206 with Content-Length ==
+    // the requested span. A 200 (full object) or a mismatched length would
+    // land the wrong bytes silently; fail the fetch and drop the
+    // connection so the retry starts clean.
+    auto cl = rh.find("Content-Length");
+    if (status != 206 || cl == rh.end() ||
+        atoll(cl->second.c_str()) != static_cast<long long>(r.length)) {
       conn.close_fd();
       return false;
     }
     // bulk MSG_WAITALL read straight into the pinned slot
-    return conn.read_body_exact(dst, r.length);
+    if (!conn.read_body_exact(dst, r.length)) {
+      conn.close_fd();
+      return false;
+    }
+    return true;
   }
 
   int device_;

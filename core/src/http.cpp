@@ -165,6 +165,10 @@ class Conn {
     return r;
   }
 
+  void shutdown_now() {
+    if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
+  }
+
   bool write_full(const char* data, size_t n) {
     while (n > 0) {
       ssize_t w = ::send(fd_, data, n, MSG_NOSIGNAL);
@@ -290,6 +294,11 @@ bool ResponseWriter::sendfile(int fd, int64_t offset, int64_t count) {
     return false;
   }
   return true;
+}
+
+void ResponseWriter::abort_connection() {
+  failed_ = true;
+  conn_->shutdown_now();
 }
 
 void ResponseWriter::write_all(int status, const std::string& body, const std::string& ct) {

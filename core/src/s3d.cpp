@@ -563,10 +563,11 @@ void handle(http::Request& req, http::ResponseWriter& w) {
     w.set_header("Accept-Ranges", "bytes");
     w.begin(status, length);
     if (req.method == "GET") {
+      bool delivered = false;
       if (!is_parts) {
         int fd = ::open(path.c_str(), O_RDONLY | O_CLOEXEC);
         if (fd >= 0) {
-          w.sendfile(fd, start, length);
+          delivered = w.sendfile(fd, start, length);
           ::close(fd);
         }
       } else {
@@ -590,7 +591,12 @@ void handle(http::Request& req, http::ResponseWriter& w) {
           remaining -= take;
           pos += psize;
         }
+        delivered = remaining == 0;
       }
+      // promised Content-Length but couldn't deliver: the stream is
+      // desynced — close the connection so the client fails loudly and
+      // retries, instead of reading the next response as body bytes
+      if (!delivered) w.abort_connection();
     }
     return;
   }
