@@ -89,9 +89,15 @@ class ResponseWriter {
 
 using Handler = std::function<void(Request&, ResponseWriter&)>;
 
+// TLS serving (reference: pkg/registry/server.go:37-43 ListenAndServeTLS).
+struct TlsConfig {
+  std::string cert_file, key_file;
+  bool enabled() const { return !cert_file.empty(); }
+};
+
 class Server {
  public:
-  Server(std::string listen_addr, Handler handler);
+  Server(std::string listen_addr, Handler handler, TlsConfig tls = {});
   ~Server();
   // binds + starts accept thread; returns bound port (for :0)
   int start();
@@ -104,6 +110,8 @@ class Server {
 
   std::string listen_addr_;
   Handler handler_;
+  TlsConfig tls_;
+  void* ssl_ctx_ = nullptr;  // SSL_CTX* when TLS is enabled
   int listen_fd_ = -1;
   int port_ = 0;
   std::atomic<bool> stopping_{false};

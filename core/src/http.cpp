@@ -1,5 +1,10 @@
 #include "modelx/http.hpp"
 
+#include <openssl/err.h>
+#include <openssl/ssl.h>
+
+#include <vector>
+
 #include <arpa/inet.h>
 #include <netdb.h>
 #include <netinet/in.h>
@@ -130,9 +135,14 @@ class Conn {
  public:
   explicit Conn(int fd) : fd_(fd) {}
   ~Conn() {
+    if (ssl_) {
+      SSL_shutdown(static_cast<SSL*>(ssl_));
+      SSL_free(static_cast<SSL*>(ssl_));
+    }
     if (fd_ >= 0) ::close(fd_);
   }
   int fd() const { return fd_; }
+  void set_ssl(void* ssl) { ssl_ = ssl; }  // takes ownership
 
   // buffered read of one line (ending \n); returns false on EOF/error
   bool read_line(std::string* line, size_t max = 65536) {
@@ -158,11 +168,7 @@ class Conn {
       rpos_ += take;
       return static_cast<ssize_t>(take);
     }
-    ssize_t r;
-    do {
-      r = ::recv(fd_, buf, n, 0);
-    } while (r < 0 && errno == EINTR);
-    return r;
+    return raw_recv(buf, n);
   }
 
   void shutdown_now() {
@@ -170,6 +176,16 @@ class Conn {
   }
 
   bool write_full(const char* data, size_t n) {
+    if (ssl_) {
+      SSL* ssl = static_cast<SSL*>(ssl_);
+      while (n > 0) {
+        int w = SSL_write(ssl, data, static_cast<int>(std::min<size_t>(n, 1u << 30)));
+        if (w <= 0) return false;
+        data += w;
+        n -= static_cast<size_t>(w);
+      }
+      return true;
+    }
     while (n > 0) {
       ssize_t w = ::send(fd_, data, n, MSG_NOSIGNAL);
       if (w < 0) {
@@ -183,6 +199,19 @@ class Conn {
   }
 
   bool sendfile_full(int in_fd, int64_t offset, int64_t count) {
+    if (ssl_) {
+      // no zero-copy under TLS: pread + SSL_write
+      std::vector<char> buf(256 << 10);
+      while (count > 0) {
+        ssize_t r = ::pread(in_fd, buf.data(),
+                            static_cast<size_t>(std::min<int64_t>(count, buf.size())), offset);
+        if (r <= 0) return false;
+        if (!write_full(buf.data(), static_cast<size_t>(r))) return false;
+        offset += r;
+        count -= r;
+      }
+      return true;
+    }
     off_t off = offset;
     while (count > 0) {
       ssize_t w = ::sendfile(fd_, in_fd, &off, static_cast<size_t>(count));
@@ -197,18 +226,32 @@ class Conn {
   }
 
  private:
-  bool fill() {
-    char buf[16384];
+  ssize_t raw_recv(char* buf, size_t n) {
+    if (ssl_) {
+      int r = SSL_read(static_cast<SSL*>(ssl_), buf,
+                       static_cast<int>(std::min<size_t>(n, 1u << 30)));
+      return r <= 0 ? (SSL_get_error(static_cast<SSL*>(ssl_), r) == SSL_ERROR_ZERO_RETURN
+                           ? 0
+                           : -1)
+                    : r;
+    }
     ssize_t r;
     do {
-      r = ::recv(fd_, buf, sizeof buf, 0);
+      r = ::recv(fd_, buf, n, 0);
     } while (r < 0 && errno == EINTR);
+    return r;
+  }
+
+  bool fill() {
+    char buf[16384];
+    ssize_t r = raw_recv(buf, sizeof buf);
     if (r <= 0) return false;
     rbuf_.assign(buf, static_cast<size_t>(r));
     rpos_ = 0;
     return true;
   }
   int fd_;
+  void* ssl_ = nullptr;  // SSL*
   std::string rbuf_;
   size_t rpos_ = 0;
 };
@@ -313,10 +356,13 @@ void ResponseWriter::write_json(int status, const std::string& body) {
 
 // --------------------------------------------------------------- server ----
 
-Server::Server(std::string listen_addr, Handler handler)
-    : listen_addr_(std::move(listen_addr)), handler_(std::move(handler)) {}
+Server::Server(std::string listen_addr, Handler handler, TlsConfig tls)
+    : listen_addr_(std::move(listen_addr)), handler_(std::move(handler)), tls_(std::move(tls)) {}
 
-Server::~Server() { stop(); }
+Server::~Server() {
+  stop();
+  if (ssl_ctx_) SSL_CTX_free(static_cast<SSL_CTX*>(ssl_ctx_));
+}
 
 int Server::start() {
   std::string host = "0.0.0.0";
@@ -326,6 +372,18 @@ int Server::start() {
     std::string h = listen_addr_.substr(0, colon);
     if (!h.empty()) host = h;
     port = atoi(listen_addr_.c_str() + colon + 1);
+  }
+  if (tls_.enabled() && !ssl_ctx_) {
+    SSL_library_init();
+    SSL_CTX* ctx = SSL_CTX_new(TLS_server_method());
+    if (!ctx) throw std::runtime_error("SSL_CTX_new failed");
+    if (SSL_CTX_use_certificate_chain_file(ctx, tls_.cert_file.c_str()) != 1 ||
+        SSL_CTX_use_PrivateKey_file(ctx, tls_.key_file.c_str(), SSL_FILETYPE_PEM) != 1 ||
+        SSL_CTX_check_private_key(ctx) != 1) {
+      SSL_CTX_free(ctx);
+      throw std::runtime_error("TLS cert/key load failed: " + tls_.cert_file);
+    }
+    ssl_ctx_ = ctx;
   }
   listen_fd_ = ::socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
   if (listen_fd_ < 0) throw std::runtime_error("socket() failed");
@@ -384,6 +442,16 @@ void Server::serve_conn(int fd, std::string peer) {
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
   Conn conn(fd);
+  if (ssl_ctx_) {
+    SSL* ssl = SSL_new(static_cast<SSL_CTX*>(ssl_ctx_));
+    if (!ssl) return;
+    SSL_set_fd(ssl, fd);
+    if (SSL_accept(ssl) != 1) {
+      SSL_free(ssl);
+      return;
+    }
+    conn.set_ssl(ssl);
+  }
   while (!stopping_.load()) {
     std::string line;
     if (!conn.read_line(&line)) return;

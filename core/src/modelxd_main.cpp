@@ -11,6 +11,7 @@
 #include <cstring>
 #include <memory>
 
+#include "modelx/http.hpp"
 #include "modelx/registry.hpp"
 #include "modelx/s3.hpp"
 
@@ -23,6 +24,7 @@ int main(int argc, char** argv) {
   bool use_s3 = false;
   bool enable_redirect = false;
   registry::AuthConfig auth;
+  http::TlsConfig tls;
 
   for (int i = 1; i < argc; i++) {
     std::string a = argv[i];
@@ -47,6 +49,8 @@ int main(int argc, char** argv) {
         pos = comma + 1;
       }
     } else if (a == "--jwt-hs256-secret") auth.jwt_hs256_secret = next();
+    else if (a == "--tls-cert") tls.cert_file = next();
+    else if (a == "--tls-key") tls.key_file = next();
     else if (a == "--help" || a == "-h") {
       printf("modelxd: modelx registry server (MI355X-native build)\n"
              "  --listen :8080            listen address\n"
@@ -57,7 +61,8 @@ int main(int argc, char** argv) {
              "  --s3-presign-expire SECS  presigned URL lifetime (default 3600)\n"
              "  --enable-redirect         hand out presigned S3 locations\n"
              "  --auth-tokens T1,T2       static bearer tokens\n"
-             "  --jwt-hs256-secret S      verify HS256 JWTs offline\n");
+             "  --jwt-hs256-secret S      verify HS256 JWTs offline\n"
+             "  --tls-cert F --tls-key F  serve HTTPS (reference --tls-*, server.go:37-43)\n");
       return 0;
     }
   }
@@ -89,7 +94,7 @@ int main(int argc, char** argv) {
     // request logging filter (reference: pkg/registry/helper.go:98-113)
     fprintf(stderr, "%s %s %lldus %s\n", req.method.c_str(), req.target.c_str(),
             static_cast<long long>(us), req.client_addr.c_str());
-  });
+  }, tls);
   int port = server.start();
   printf("modelxd listening on port %d backend=%s redirect=%d\n", port,
          use_s3 ? "s3" : "local", enable_redirect ? 1 : 0);

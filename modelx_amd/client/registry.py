@@ -12,11 +12,21 @@ from ..wire import errors as er
 from ..wire import types
 
 
+class _InsecureSession(requests.Session):
+    """requests >= 2.32 stopped honoring session-level ``verify=False`` in
+    some paths — inject it per request (the --insecure TLS skip,
+    reference: cmd/modelx/modelx.go:29-36)."""
+
+    def request(self, *args, **kwargs):  # noqa: D102
+        kwargs.setdefault("verify", False)
+        return super().request(*args, **kwargs)
+
+
 class RegistryClient:
     def __init__(self, registry: str, authorization: str = "", insecure: bool = False):
         self.registry = registry.rstrip("/")
         self.authorization = authorization
-        self.session = requests.Session()
+        self.session = _InsecureSession() if insecure else requests.Session()
         self.session.verify = not insecure
         self.user_agent = f"modelx/{__version__}"
 
