@@ -92,8 +92,14 @@ class TestConcurrentPulls:
                 ref[f"d{i}"] = t
             for it in range(2):
                 client.clear_chunk_index()
-                outs = client.pull_many("conc/dd", list(ref), parallel=6)
+                # prime the index with one pull, then hit it from 5 threads
+                prime = client.pull_to_gpu("conc/dd", "d0")
+                assert torch.equal(prime["blob.bin"], ref["d0"]), it
+                outs = client.pull_many("conc/dd", [f"d{i}" for i in range(1, 6)],
+                                        parallel=5)
                 for v, t in ref.items():
+                    if v == "d0":
+                        continue
                     assert torch.equal(outs[v]["blob.bin"], t), (it, v)
             dd = [s for s in client.last_stats if s.get("phase") == "pull-dedup"]
             assert dd, "dedup path never taken"
