@@ -282,11 +282,21 @@ class GpuClient:
                     t0 = time.monotonic()
                     self.engine.tar_scatter(0, segs)
                     fetched = self._fetch_ranges(url, headers, tensor.data_ptr(), missing)
-                    self.last_stats.append({
-                        "phase": "pull-dedup", "bytes": fetched,
-                        "dedup_bytes": sum(s[2] for s in segs),
-                        "seconds": time.monotonic() - t0})
+                    stats = {"phase": "pull-dedup", "bytes": fetched,
+                             "dedup_bytes": sum(s[2] for s in segs),
+                             "seconds": time.monotonic() - t0}
+                    self.last_stats.append(stats)
                     if verify:
+                        # verify against the leaves; refetch any bad chunk
+                        # (a stale/poisoned index entry must degrade to a
+                        # fetch, never to a failure)
+                        got = self.engine.sha256_chunk_leaves(tensor.data_ptr(),
+                                                              desc.size, cs)
+                        bad = self._bad_chunk_ranges(got, expect, cs, desc.size)
+                        if bad:
+                            stats["refetched_bytes"] = self._fetch_ranges(
+                                url, headers, tensor.data_ptr(), bad)
+                            stats["refetched_ranges"] = len(bad)
                         self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
                     self.register_chunks(tensor, expect, cs)
                     return tensor
