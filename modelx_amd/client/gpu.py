@@ -294,10 +294,30 @@ class GpuClient:
                                                               desc.size, cs)
                         bad = self._bad_chunk_ranges(got, expect, cs, desc.size)
                         if bad:
+                            import sys
+
+                            gathered = {s[1] - tensor.data_ptr() for s in segs}
+                            kinds = {"gathered": 0, "fetched": 0}
+                            for off, ln in bad:
+                                for o in range(off, off + ln, cs):
+                                    kinds["gathered" if o in gathered else "fetched"] += 1
+                            print(f"modelx: dedup refetch {desc.name}: "
+                                  f"{len(bad)} ranges, chunks by origin {kinds}",
+                                  file=sys.stderr)
                             stats["refetched_bytes"] = self._fetch_ranges(
                                 url, headers, tensor.data_ptr(), bad)
                             stats["refetched_ranges"] = len(bad)
-                        self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+                            got = self.engine.sha256_chunk_leaves(tensor.data_ptr(),
+                                                                  desc.size, cs)
+                        root = dg.root_from_leaf_bytes(got, cs, desc.size)
+                        target = desc.annotations.get(types.ANNOTATION_CHUNK_DIGEST, "")
+                        if dg.algo_chunk_size(desc.digest.split(":", 1)[0] if desc.digest
+                                              else ""):
+                            target = desc.digest
+                        if target and root != target:
+                            raise er.ModelxError(
+                                er.ErrCode.DIGEST_INVALID,
+                                f"GPU chunk digest mismatch for {desc.name}: {root}")
                     self.register_chunks(tensor, expect, cs)
                     return tensor
 
