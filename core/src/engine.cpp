@@ -57,6 +57,18 @@ extern "C" hipError_t modelx_zstd_decompress_frames(const void* src, const void*
                                                     uint32_t nframes, void* dst,
                                                     void* lit_scratch, int64_t* rc_dev,
                                                     hipStream_t stream);
+extern "C" hipError_t modelx_dedup_insert(const void* leaves_dev, uint32_t nchunks,
+                                           uint64_t base, uint64_t chunk_size, uint64_t total,
+                                           void* table, uint64_t cap, uint32_t* dropped_dev,
+                                           hipStream_t stream);
+extern "C" hipError_t modelx_dedup_probe(const void* leaves_dev, uint32_t nchunks,
+                                         uint64_t chunk_size, uint64_t total, const void* table,
+                                         uint64_t cap, uint64_t* src_addr_dev,
+                                         uint64_t* src_len_dev, hipStream_t stream);
+extern "C" hipError_t modelx_dedup_gather(const uint64_t* src_addr_dev,
+                                          const uint64_t* src_len_dev, uint32_t nchunks,
+                                          uint64_t dst_base, uint64_t chunk_size,
+                                          hipStream_t stream);
 extern "C" hipError_t modelx_zstd_compress_frames(const void* src, uint64_t srclen,
                                                   uint32_t frame_raw, uint32_t first_frame,
                                                   uint32_t nframes, void* dst_scratch,
@@ -120,6 +132,10 @@ class GpuEngine {
   }
 
   ~GpuEngine() {
+    if (dedup_table_) hipFree(dedup_table_);
+    if (dedup_dropped_) hipFree(dedup_dropped_);
+    for (auto& p : ds_ptr_)
+      if (p) hipFree(p);
     for (auto& p : zs_ptr_)
       if (p) hipFree(p);
     for (auto& st : streams_) hipStreamDestroy(st);
@@ -617,6 +633,17 @@ class GpuEngine {
     return zs_ptr_[idx];
   }
 
+  void* dedup_scratch(size_t idx, size_t need) {
+    if (ds_size_[idx] < need) {
+      if (ds_ptr_[idx]) HIP_CHECK(hipFree(ds_ptr_[idx]));
+      ds_ptr_[idx] = nullptr;
+      ds_size_[idx] = 0;
+      HIP_CHECK(hipMalloc(&ds_ptr_[idx], need));
+      ds_size_[idx] = need;
+    }
+    return ds_ptr_[idx];
+  }
+
   uint64_t zstd_compress_device(uintptr_t src_ptr, uint64_t size, uint32_t frame_raw,
                                 uintptr_t dst_ptr, uint64_t dst_cap) {
     HIP_CHECK(hipSetDevice(device_));
@@ -741,6 +768,101 @@ class GpuEngine {
     return nframes * stride + 8 + nframes * 8 + 9;
   }
 
+  // ------------------------------------------------------ chunk dedup ----
+  // HBM-resident chunk hash table (core/hip/dedup.hip). Insert/probe/gather
+  // run on hash_stream_ under dedup_mu_; the caller (GpuClient) keeps the
+  // registered tensors alive.
+
+  void dedup_reset(uint64_t cap_pow2) {
+    HIP_CHECK(hipSetDevice(device_));
+    py::gil_scoped_release release;
+    std::lock_guard<std::mutex> lk(dedup_mu_);
+    if (cap_pow2 == 0) cap_pow2 = dedup_cap_ ? dedup_cap_ : (1ull << 22);
+    if (dedup_cap_ != cap_pow2) {
+      if (dedup_table_) HIP_CHECK(hipFree(dedup_table_));
+      dedup_table_ = nullptr;
+      HIP_CHECK(hipMalloc(&dedup_table_, cap_pow2 * 48));
+      dedup_cap_ = cap_pow2;
+    }
+    if (!dedup_dropped_)
+      HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dedup_dropped_), sizeof(uint32_t)));
+    HIP_CHECK(hipMemsetAsync(dedup_table_, 0, dedup_cap_ * 48, hash_stream_));
+    HIP_CHECK(hipMemsetAsync(dedup_dropped_, 0, sizeof(uint32_t), hash_stream_));
+    HIP_CHECK(hipStreamSynchronize(hash_stream_));
+  }
+
+  // Register a blob's chunks (leaves = packed 32 B digests, host bytes).
+  // Returns the running dropped-entry count (table pressure indicator).
+  uint32_t dedup_register(py::bytes leaves, uintptr_t base, uint64_t chunk_size,
+                          uint64_t total) {
+    std::string lv = leaves;
+    HIP_CHECK(hipSetDevice(device_));
+    py::gil_scoped_release release;
+    std::lock_guard<std::mutex> lk(dedup_mu_);
+    if (!dedup_table_) {
+      // lazy init at default capacity
+      uint64_t cap = 1ull << 22;
+      HIP_CHECK(hipMalloc(&dedup_table_, cap * 48));
+      HIP_CHECK(hipMemsetAsync(dedup_table_, 0, cap * 48, hash_stream_));
+      dedup_cap_ = cap;
+      HIP_CHECK(hipMalloc(reinterpret_cast<void**>(&dedup_dropped_), sizeof(uint32_t)));
+      HIP_CHECK(hipMemsetAsync(dedup_dropped_, 0, sizeof(uint32_t), hash_stream_));
+    }
+    uint32_t n = static_cast<uint32_t>(lv.size() / 32);
+    void* dleaves = dedup_scratch(0, lv.size() ? lv.size() : 32);
+    HIP_CHECK(hipMemcpyAsync(dleaves, lv.data(), lv.size(), hipMemcpyHostToDevice,
+                             hash_stream_));
+    HIP_CHECK(modelx_dedup_insert(dleaves, n, base, chunk_size, total, dedup_table_,
+                                  dedup_cap_, dedup_dropped_, hash_stream_));
+    uint32_t dropped = 0;
+    HIP_CHECK(hipStreamSynchronize(hash_stream_));
+    HIP_CHECK(hipMemcpy(&dropped, dedup_dropped_, sizeof dropped, hipMemcpyDeviceToHost));
+    return dropped;
+  }
+
+  // Probe + gather resident chunks for an expected-leaves array; returns
+  // (missing (offset,length) ranges merged, deduped_bytes).
+  py::tuple dedup_pull(py::bytes expect, uintptr_t dst_ptr, uint64_t chunk_size,
+                       uint64_t total) {
+    std::string lv = expect;
+    uint32_t n = static_cast<uint32_t>(lv.size() / 32);
+    std::vector<std::pair<uint64_t, uint64_t>> missing;
+    uint64_t deduped = 0;
+    if (n && dedup_table_) {
+      HIP_CHECK(hipSetDevice(device_));
+      py::gil_scoped_release release;
+      std::lock_guard<std::mutex> lk(dedup_mu_);
+      void* dleaves = dedup_scratch(0, lv.size());
+      uint64_t* daddr = reinterpret_cast<uint64_t*>(dedup_scratch(1, (uint64_t)n * 8));
+      uint64_t* dlen = reinterpret_cast<uint64_t*>(dedup_scratch(2, (uint64_t)n * 8));
+      HIP_CHECK(hipMemcpyAsync(dleaves, lv.data(), lv.size(), hipMemcpyHostToDevice,
+                               hash_stream_));
+      HIP_CHECK(modelx_dedup_probe(dleaves, n, chunk_size, total, dedup_table_, dedup_cap_,
+                                   daddr, dlen, hash_stream_));
+      HIP_CHECK(modelx_dedup_gather(daddr, dlen, n, dst_ptr, chunk_size, hash_stream_));
+      std::vector<uint64_t> haddr(n);
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+      HIP_CHECK(hipMemcpy(haddr.data(), daddr, (uint64_t)n * 8, hipMemcpyDeviceToHost));
+      for (uint32_t i = 0; i < n; i++) {
+        uint64_t off = (uint64_t)i * chunk_size;
+        uint64_t ln = total - off < chunk_size ? total - off : chunk_size;
+        if (ln == 0) break;
+        if (haddr[i] != 0) {
+          deduped += ln;
+        } else if (!missing.empty() && missing.back().first + missing.back().second == off) {
+          missing.back().second += ln;
+        } else {
+          missing.emplace_back(off, ln);
+        }
+      }
+    } else {
+      if (total) missing.emplace_back(0, total);
+    }
+    py::list out;
+    for (auto& m : missing) out.append(py::make_tuple(m.first, m.second));
+    return py::make_tuple(out, deduped);
+  }
+
   void synchronize() {
     HIP_CHECK(hipSetDevice(device_));
     for (auto& st : streams_) HIP_CHECK(hipStreamSynchronize(st));
@@ -803,6 +925,12 @@ class GpuEngine {
   std::mutex zstd_mu_;
   void* zs_ptr_[7] = {};
   size_t zs_size_[7] = {};
+  std::mutex dedup_mu_;
+  void* dedup_table_ = nullptr;
+  uint64_t dedup_cap_ = 0;
+  uint32_t* dedup_dropped_ = nullptr;
+  void* ds_ptr_[3] = {};  // dedup scratch: leaves, addr, len (under dedup_mu_)
+  size_t ds_size_[3] = {};
 };
 
 bool hip_available() {
@@ -841,6 +969,11 @@ PYBIND11_MODULE(_core, m) {
            py::arg("method"), py::arg("headers"), py::arg("src_ptr"), py::arg("size"))
       .def("tar_index", &GpuEngine::tar_index, py::arg("tar_ptr"), py::arg("tar_len"))
       .def("tar_scatter", &GpuEngine::tar_scatter, py::arg("tar_ptr"), py::arg("segs"))
+      .def("dedup_reset", &GpuEngine::dedup_reset, py::arg("cap_pow2") = (uint64_t)0)
+      .def("dedup_register", &GpuEngine::dedup_register, py::arg("leaves"), py::arg("base"),
+           py::arg("chunk_size"), py::arg("total"))
+      .def("dedup_pull", &GpuEngine::dedup_pull, py::arg("expect"), py::arg("dst_ptr"),
+           py::arg("chunk_size"), py::arg("total"))
       .def("zstd_compress_device", &GpuEngine::zstd_compress_device, py::arg("src_ptr"),
            py::arg("size"), py::arg("frame_raw"), py::arg("dst_ptr"), py::arg("dst_cap"))
       .def("zstd_decompress_device", &GpuEngine::zstd_decompress_device, py::arg("src_ptr"),

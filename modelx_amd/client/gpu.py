@@ -73,13 +73,14 @@ class GpuClient:
         self.engine = core.GpuEngine(device=device, num_slots=num_slots,
                                      slot_bytes=slot_bytes, num_streams=4)
         self.last_stats: List[dict] = []
-        # cross-blob chunk dedup (SURVEY.md §2.2 chunk_verify_dedup): leaf
-        # digest -> (tensor, offset, length) of an HBM-resident chunk. When a
-        # pull's expected leaves hit the index, the chunk is gathered D2D at
-        # HBM bandwidth (~8 TB/s) instead of re-fetched from S3. Opt-in: the
-        # index holds tensor references (keeps them alive).
+        # cross-blob chunk dedup (SURVEY.md §2.2 chunk_verify_dedup): an
+        # HBM-resident hash table (core/hip/dedup.hip) maps leaf digests to
+        # resident chunk addresses; a pull probes it on-device and gathers
+        # hits D2D at HBM bandwidth instead of re-fetching from S3. Opt-in:
+        # registered tensors are kept alive in _dedup_tensors.
         self.dedup = dedup
-        self._chunk_index: Dict[bytes, Tuple[object, int, int]] = {}
+        self._dedup_tensors: List[object] = []
+        self._dedup_registered = False
 
     # ----------------------------------------------------------- helpers --
 
@@ -156,43 +157,20 @@ class GpuClient:
     # ------------------------------------------------------ chunk dedup --
 
     def register_chunks(self, tensor, leaves: bytes, chunk_size: int) -> None:
-        """Add a blob's chunks to the dedup index (leaf digest -> location)."""
+        """Add a blob's chunks to the HBM dedup table (dedup.hip insert
+        kernel); keeps the tensor alive while registered."""
         if not self.dedup:
             return
         size = tensor.numel() * tensor.element_size()
-        n = len(leaves) // 32
-        for i in range(n):
-            off = i * chunk_size
-            ln = min(chunk_size, size - off)
-            if ln <= 0:
-                break
-            self._chunk_index.setdefault(leaves[i * 32 : (i + 1) * 32], (tensor, off, ln))
+        self._dedup_tensors.append(tensor)
+        self.engine.dedup_register(leaves, tensor.data_ptr(), chunk_size, size)
+        self._dedup_registered = True
 
     def clear_chunk_index(self) -> None:
-        self._chunk_index.clear()
-
-    def _dedup_plan(self, expect: bytes, chunk_size: int, size: int,
-                    dst_ptr: int) -> Tuple[List[Tuple[int, int, int]], List[Tuple[int, int]]]:
-        """Split a pull into (D2D gather segs for resident chunks, byte
-        ranges to fetch). Gather segs are absolute (src_addr, dst_addr, len)
-        for the generic copy kernel."""
-        segs: List[Tuple[int, int, int]] = []
-        missing: List[Tuple[int, int]] = []
-        n = len(expect) // 32
-        for i in range(n):
-            off = i * chunk_size
-            ln = min(chunk_size, size - off)
-            if ln <= 0:
-                break
-            hit = self._chunk_index.get(expect[i * 32 : (i + 1) * 32])
-            if hit is not None and hit[2] == ln:
-                t, src_off, _ = hit
-                segs.append((t.data_ptr() + src_off, dst_ptr + off, ln))
-            elif missing and missing[-1][0] + missing[-1][1] == off:
-                missing[-1] = (missing[-1][0], missing[-1][1] + ln)
-            else:
-                missing.append((off, ln))
-        return segs, missing
+        if self._dedup_registered:
+            self.engine.dedup_reset(0)
+        self._dedup_tensors.clear()
+        self._dedup_registered = False
 
     def _fetch_ranges(self, url: str, headers: Dict[str, str], ptr: int,
                       ranges: List[Tuple[int, int]]) -> int:
@@ -271,19 +249,20 @@ class GpuClient:
                 return tensor
             # no leaves sidecar → fall through to a full pull
 
-        if self.dedup and self._chunk_index:
-            # cross-blob dedup: gather HBM-resident chunks D2D, fetch only
-            # the rest (SURVEY.md §2.2 chunk_verify_dedup; the reference
-            # dedups at whole-blob granularity only, push.go:169-177)
+        if self.dedup and self._dedup_registered:
+            # cross-blob dedup: probe the HBM table and gather resident
+            # chunks D2D, fetch only the rest (SURVEY.md §2.2
+            # chunk_verify_dedup; the reference dedups at whole-blob
+            # granularity only, push.go:169-177)
             expect = self._expected_leaves(repository, desc)
             if expect is not None:
-                segs, missing = self._dedup_plan(expect, cs, desc.size, tensor.data_ptr())
-                if segs:
-                    t0 = time.monotonic()
-                    self.engine.tar_scatter(0, segs)
+                t0 = time.monotonic()
+                missing, dedup_bytes = self.engine.dedup_pull(
+                    expect, tensor.data_ptr(), cs, desc.size)
+                if dedup_bytes:
                     fetched = self._fetch_ranges(url, headers, tensor.data_ptr(), missing)
                     stats = {"phase": "pull-dedup", "bytes": fetched,
-                             "dedup_bytes": sum(s[2] for s in segs),
+                             "dedup_bytes": dedup_bytes,
                              "seconds": time.monotonic() - t0}
                     self.last_stats.append(stats)
                     if verify:
@@ -296,11 +275,14 @@ class GpuClient:
                         if bad:
                             import sys
 
-                            gathered = {s[1] - tensor.data_ptr() for s in segs}
+                            fetched_offs = set()
+                            for off, ln in missing:
+                                for o in range(off, off + ln, cs):
+                                    fetched_offs.add(o)
                             kinds = {"gathered": 0, "fetched": 0}
                             for off, ln in bad:
                                 for o in range(off, off + ln, cs):
-                                    kinds["gathered" if o in gathered else "fetched"] += 1
+                                    kinds["fetched" if o in fetched_offs else "gathered"] += 1
                             print(f"modelx: dedup refetch {desc.name}: "
                                   f"{len(bad)} ranges, chunks by origin {kinds}",
                                   file=sys.stderr)

@@ -18,6 +18,7 @@ SOURCES = [
     "core/src/engine.cpp",
     "core/hip/sha256.hip",
     "core/hip/tar.hip",
+    "core/hip/dedup.hip",
     "core/hip/zstd.hip",
     "core/src/zstd_cpu.cpp",
     "core/src/http.cpp",

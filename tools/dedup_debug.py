@@ -39,7 +39,7 @@ def main():
 
         out0 = g.pull_to_gpu("dbg/dd", "d0")
         assert torch.equal(out0["blob.bin"], blobs["d0"]), "d0 plain pull wrong"
-        print("d0 ok; chunk index:", len(g._chunk_index))
+        print("d0 ok; dedup tensors held:", len(g._dedup_tensors))
 
         # manual replay of the dedup path for d1 with full introspection
         man = g.remote.get_manifest("dbg/dd", "d1")
@@ -47,19 +47,21 @@ def main():
         expect = g._expected_leaves("dbg/dd", desc)
         assert expect is not None, "no leaves sidecar"
         dst = torch.empty(desc.size, dtype=torch.uint8, device="cuda:0")
-        segs, missing = g._dedup_plan(expect, CS, desc.size, dst.data_ptr())
-        print(f"plan: {len(segs)} dedup segs, missing ranges {missing}")
+        missing, dedup_bytes = g.engine.dedup_pull(expect, dst.data_ptr(), CS, desc.size)
+        print(f"plan: {dedup_bytes} deduped bytes, missing ranges {missing}")
         url, headers = g._download_url("dbg/dd", desc)
-        g.engine.tar_scatter(0, segs)
         fetched = g._fetch_ranges(url, headers, dst.data_ptr(), missing)
         print("fetched", fetched)
         got = g.engine.sha256_chunk_leaves(dst.data_ptr(), desc.size, CS)
         nchunks = len(expect) // 32
-        dedup_dst_offs = {s[1] - dst.data_ptr() for s in segs}
+        fetched_offs = set()
+        for off, ln in missing:
+            for o in range(off, off + ln, CS):
+                fetched_offs.add(o)
         bad = []
         for i in range(nchunks):
             if got[i * 32 : (i + 1) * 32] != expect[i * 32 : (i + 1) * 32]:
-                kind = "GATHERED" if i * CS in dedup_dst_offs else "FETCHED"
+                kind = "FETCHED" if i * CS in fetched_offs else "GATHERED"
                 bad.append((i, kind))
         print(f"bad chunks: {len(bad)} of {nchunks}")
         for i, kind in bad[:20]:
