@@ -66,7 +66,7 @@ def build() -> str:
         print("+", " ".join(cmd), flush=True)
         subprocess.run(cmd, check=True)
         objs.append(obj)
-    link = [HIPCC, "-shared", "-fPIC", "-o", out] + objs + ["-lcrypto"]
+    link = [HIPCC, "-shared", "-fPIC", "-o", out] + objs + ["-lcrypto", "-lssl"]
     print("+", " ".join(link), flush=True)
     subprocess.run(link, check=True)
     return out
