@@ -3,6 +3,7 @@ Same endpoints, headers and error decoding; User-Agent ``modelx/<ver>``."""
 from __future__ import annotations
 
 import json
+import threading
 from typing import Any, Dict, Iterator, Optional
 
 import requests
@@ -22,12 +23,34 @@ class _InsecureSession(requests.Session):
         return super().request(*args, **kwargs)
 
 
+class _SessionPerThread:
+    """requests.Session is not thread-safe (psf/requests#1871): a shared
+    session under concurrent use (GpuClient.pull_many) can cross responses
+    between threads — observed on hardware as a pull landing ANOTHER blob's
+    presigned content (self-consistent wrong bytes, digest mismatch on
+    every chunk). One session per thread, same API surface."""
+
+    def __init__(self, insecure: bool):
+        self._insecure = insecure
+        self._local = threading.local()
+
+    def _s(self) -> requests.Session:
+        s = getattr(self._local, "s", None)
+        if s is None:
+            s = _InsecureSession() if self._insecure else requests.Session()
+            s.verify = not self._insecure
+            self._local.s = s
+        return s
+
+    def __getattr__(self, name):
+        return getattr(self._s(), name)
+
+
 class RegistryClient:
     def __init__(self, registry: str, authorization: str = "", insecure: bool = False):
         self.registry = registry.rstrip("/")
         self.authorization = authorization
-        self.session = _InsecureSession() if insecure else requests.Session()
-        self.session.verify = not insecure
+        self.session = _SessionPerThread(insecure)
         self.user_agent = f"modelx/{__version__}"
 
     # --- internals -------------------------------------------------------
