@@ -1,6 +1,8 @@
 #!/usr/bin/env python3
-"""A/B the object-store backing: plain tmpfs vs transparent-hugepage tmpfs,
-PUT-then-immediately-GET (the bench pattern). Requires root (mounts tmpfs)."""
+"""A/B the object-store backing, PUT-then-immediately-GET (the bench
+pattern). Mount-free variants (gpurun boxes refuse mounts): existing
+/dev/shm (tmpfs) vs the scratch disk. Pass a mount-opts string as a third
+variant only where mounting is permitted."""
 import os
 import subprocess
 import sys
@@ -18,15 +20,18 @@ from util_servers import S3D, free_port, wait_http
 SIZE = 8 << 30
 
 
-def bench_store(tag, mount_opts):
-    root = f"/mnt/modelx-{tag}"
-    os.makedirs(root, exist_ok=True)
-    subprocess.run(["umount", root], capture_output=True)
-    r = subprocess.run(["mount", "-t", "tmpfs", "-o", mount_opts, "none", root],
-                       capture_output=True, text=True)
-    if r.returncode != 0:
-        print(f"{tag}: mount failed: {r.stderr.strip()}")
-        return
+def bench_store(tag, root_base=None, mount_opts=None):
+    if mount_opts:
+        root = f"/mnt/modelx-{tag}"
+        os.makedirs(root, exist_ok=True)
+        subprocess.run(["umount", root], capture_output=True)
+        r = subprocess.run(["mount", "-t", "tmpfs", "-o", mount_opts, "none", root],
+                           capture_output=True, text=True)
+        if r.returncode != 0:
+            print(f"{tag}: mount failed ({r.stderr.strip()}) — skipping")
+            return
+    else:
+        root = os.path.join(root_base, f"modelx-{tag}")
     os.makedirs(f"{root}/modelx", exist_ok=True)
     port = free_port()
     proc = subprocess.Popen([S3D, "--listen", f"127.0.0.1:{port}", "--root", root, "--no-auth"],
@@ -61,9 +66,15 @@ def bench_store(tag, mount_opts):
     finally:
         proc.terminate()
         proc.wait()
-        subprocess.run(["umount", root], capture_output=True)
+        if mount_opts:
+            subprocess.run(["umount", root], capture_output=True)
+        else:
+            import shutil
+
+            shutil.rmtree(root, ignore_errors=True)
 
 
 if __name__ == "__main__":
-    bench_store("plain", "size=48g")
-    bench_store("huge", "size=48g,huge=always")
+    bench_store("shm", root_base="/dev/shm")
+    bench_store("disk", root_base="/tmp")
+    bench_store("hugetmpfs", mount_opts="size=48g,huge=always")
