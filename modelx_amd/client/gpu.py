@@ -338,19 +338,34 @@ class GpuClient:
         t0 = time.monotonic()
         try:
             self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
-        except er.ModelxError:
+        except er.ModelxError as first_err:
             # chunk-level refetch before giving up
             expect = self._expected_leaves(repository, desc)
             if expect is None:
-                raise
+                raise er.ModelxError(
+                    er.ErrCode.DIGEST_INVALID,
+                    f"{first_err.message} [size={desc.size} media={desc.media_type} "
+                    f"path=no-leaves-sidecar]") from first_err
             got = self.engine.sha256_chunk_leaves(tensor.data_ptr(), desc.size, cs)
             ranges = self._bad_chunk_ranges(got, expect, cs, desc.size)
             if not ranges:
-                raise
+                # landed leaves MATCH the sidecar yet the root digest
+                # differs: the sidecar and the descriptor disagree — a
+                # metadata-level inconsistency, not a transfer corruption
+                raise er.ModelxError(
+                    er.ErrCode.DIGEST_INVALID,
+                    f"{first_err.message} [size={desc.size} media={desc.media_type} "
+                    f"path=leaves-match-but-root-differs]") from first_err
             self._fetch_ranges(url, headers, tensor.data_ptr(), ranges)
             self.last_stats.append({"phase": "pull-chunk-refetch",
                                     "bytes": sum(r[1] for r in ranges)})
-            self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+            try:
+                self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+            except er.ModelxError as second_err:
+                raise er.ModelxError(
+                    er.ErrCode.DIGEST_INVALID,
+                    f"{second_err.message} [size={desc.size} media={desc.media_type} "
+                    f"path=refetch-did-not-fix ranges={len(ranges)}]") from second_err
         self.last_stats.append({"phase": "pull-verify", "bytes": desc.size,
                                 "seconds": time.monotonic() - t0})
         return tensor
