@@ -84,6 +84,19 @@ class GpuClient:
 
     # ----------------------------------------------------------- helpers --
 
+    def _sync_producers(self) -> None:
+        """Synchronize torch's streams before the engine READS a
+        user-provided tensor. The engine runs on its own non-blocking HIP
+        streams, which never implicitly order against torch's default
+        stream — without this, digesting/pushing a tensor whose producing
+        kernel (randint/repeat/copy) is still in flight reads torn bytes
+        and permanently stores content that matches no digest. (Observed
+        on hardware with 2.4 GiB generated tensors; small tensors usually
+        win the race, which is what made it look intermittent.)"""
+        import torch
+
+        torch.cuda.synchronize(self.device)
+
     def _verify_device_digest(self, ptr: int, size: int, desc: types.Descriptor) -> None:
         """GPU chunk-digest the landed buffer and compare against the
         descriptor's chunked digest (or its chunk annotation)."""
@@ -460,6 +473,7 @@ class GpuClient:
                                        ) -> Tuple[str, bytes]:
         import time
 
+        self._sync_producers()
         t0 = time.monotonic()
         leaves = self.engine.sha256_chunk_leaves(ptr, size, chunk_size)
         root = dg.root_from_leaf_bytes(leaves, chunk_size, size)
@@ -485,6 +499,7 @@ class GpuClient:
         (push.go:169-177 semantics)."""
         import time
 
+        self._sync_producers()
         t0 = time.monotonic()
         if self.remote.head_blob(repository, desc.digest):
             return
@@ -555,6 +570,7 @@ class GpuClient:
             comp_keep = None  # keep the compressed tensor alive until pushed
             if compress == "zstd" and size:
                 core = _core()
+                self._sync_producers()
                 raw_root, _ = self.digest_device_blob_with_leaves(t.data_ptr(), size,
                                                                   chunk_size)
                 bound = core.zstd_compress_bound(size)
