@@ -56,6 +56,9 @@ struct Request {
   Conn* conn = nullptr;
   int64_t body_remaining = 0;
   ssize_t read_body(char* buf, size_t n);          // returns 0 at end
+  // raw socket fd when the connection is plaintext AND the conn buffer is
+  // fully drained (returns -1 otherwise) — enables splice()-based sinks
+  int raw_fd_if_plain() const;
   std::string read_body_all(size_t max_bytes);      // throws if over max
   void drain_body();
 };

@@ -175,6 +175,9 @@ class Conn {
     if (fd_ >= 0) ::shutdown(fd_, SHUT_RDWR);
   }
 
+  // raw fd for splice()-style consumers; -1 when TLS or bytes are buffered
+  int raw_fd_if_plain() const { return (!ssl_ && rpos_ >= rbuf_.size()) ? fd_ : -1; }
+
   bool write_full(const char* data, size_t n) {
     if (ssl_) {
       SSL* ssl = static_cast<SSL*>(ssl_);
@@ -257,6 +260,8 @@ class Conn {
 };
 
 // -------------------------------------------------------------- request ----
+
+int Request::raw_fd_if_plain() const { return conn ? conn->raw_fd_if_plain() : -1; }
 
 ssize_t Request::read_body(char* buf, size_t n) {
   if (body_remaining <= 0) return 0;

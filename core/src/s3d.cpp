@@ -91,6 +91,68 @@ bool write_stream_to(const std::string& path, http::Request& req, std::string* e
   // mapping) was A/B-tested on the GPU box and LOST to plain write() on the
   // tmpfs store: per-page fault handling in the recv copy path costs more
   // than the buffer bounce it saves (push throughput 7.5 vs 12.5 GiB/s).
+  // MODELX_S3D_SPLICE=1 enables the splice(socket->pipe->file) sink for
+  // A/B (zero user-space copies; tools/store_micro.py measures it).
+  static const bool use_splice = [] {
+    const char* v = getenv("MODELX_S3D_SPLICE");
+    return v && *v == '1';
+  }();
+  if (use_splice && req.content_length > 0) {
+    // drain any bytes the header parser buffered first
+    char small[16384];
+    while (req.body_remaining > 0 && req.raw_fd_if_plain() < 0) {
+      ssize_t r = req.read_body(small, sizeof small);
+      if (r <= 0) break;
+      ssize_t off = 0;
+      while (off < r) {
+        ssize_t w = ::write(fd, small + off, static_cast<size_t>(r - off));
+        if (w < 0 && errno == EINTR) continue;
+        if (w < 0) break;
+        off += w;
+      }
+      total += r;
+    }
+    int sock = req.raw_fd_if_plain();
+    if (sock >= 0 && req.body_remaining > 0) {
+      int pfd[2];
+      if (::pipe(pfd) == 0) {
+        fcntl(pfd[0], F_SETPIPE_SZ, 1 << 20);
+        int64_t rem = req.body_remaining;
+        while (rem > 0) {
+          ssize_t r = ::splice(sock, nullptr, pfd[1], nullptr,
+                               static_cast<size_t>(std::min<int64_t>(rem, 1 << 20)),
+                               SPLICE_F_MOVE | SPLICE_F_MORE);
+          if (r < 0 && errno == EINTR) continue;
+          if (r <= 0) break;
+          ssize_t moved = 0;
+          while (moved < r) {
+            ssize_t w = ::splice(pfd[0], nullptr, fd, nullptr,
+                                 static_cast<size_t>(r - moved), SPLICE_F_MOVE);
+            if (w < 0 && errno == EINTR) continue;
+            if (w <= 0) { moved = -1; break; }
+            moved += w;
+          }
+          if (moved < 0) break;
+          rem -= r;
+          total += r;
+        }
+        req.body_remaining = rem;
+        ::close(pfd[0]);
+        ::close(pfd[1]);
+      }
+    }
+    ::close(fd);
+    if (total != req.content_length) {
+      ::unlink(tmp.c_str());
+      return false;
+    }
+    if (::rename(tmp.c_str(), path.c_str()) != 0) {
+      ::unlink(tmp.c_str());
+      return false;
+    }
+    if (etag) *etag = "\"s3d-" + std::to_string(total) + "\"";
+    return true;
+  }
   std::vector<char> buf(4 << 20);
   while (true) {
     ssize_t r = req.read_body(buf.data(), buf.size());
