@@ -34,9 +34,9 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--blob-gib", type=float, default=4.0,
                    help="per-rank blob size per step (GiB)")
-    p.add_argument("--conns", type=int, default=8, help="ranged-GET connections per rank")
+    p.add_argument("--conns", type=int, default=16, help="ranged-GET connections per rank")
     p.add_argument("--slot-mib", type=int, default=64)
-    p.add_argument("--slots", type=int, default=10)
+    p.add_argument("--slots", type=int, default=16)
     p.add_argument("--part-mib", type=int, default=256, help="push part size")
     p.add_argument("--store", default="", help="s3d store root (default: tmpfs)")
     p.add_argument("--cpu-smoke", action="store_true",

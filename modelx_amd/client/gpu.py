@@ -28,7 +28,7 @@ from .registry import RegistryClient
 
 DEFAULT_SLOT_BYTES = 64 << 20
 DEFAULT_NUM_SLOTS = 16
-DEFAULT_NUM_CONNS = 8
+DEFAULT_NUM_CONNS = 16  # measured: 16 conns/16 slots -> 17.3 vs 14.3 GiB/s at 8/10 (bench sweep)
 DEFAULT_PART_BYTES = 256 << 20  # push part granularity (multipart)
 DEFAULT_PUSH_PARALLEL = 6
 # 128 KiB chunks put 65k+ SHA-256 chains in flight for multi-GiB blobs — the
