@@ -30,6 +30,8 @@
 #include <condition_variable>
 #include <cstring>
 #include <deque>
+#include <map>
+#include <memory>
 #include <mutex>
 #include <queue>
 #include <stdexcept>
@@ -112,6 +114,33 @@ struct Slot {
 struct Range {
   uint64_t offset;
   uint64_t length;
+};
+
+// Keep-alive connection pool: checkout/checkin per (host,port). Reused
+// sockets skip TCP connect + slow-start on every blob of an index — the
+// per-blob latency term that dominates many-small-blob pulls (config 5).
+class ConnPool {
+ public:
+  std::unique_ptr<http::ClientConn> checkout(const std::string& host, int port) {
+    std::lock_guard<std::mutex> lk(mu_);
+    auto& v = pool_[host + ":" + std::to_string(port)];
+    if (!v.empty()) {
+      auto c = std::move(v.back());
+      v.pop_back();
+      return c;
+    }
+    return std::make_unique<http::ClientConn>(host, port);
+  }
+  void checkin(std::unique_ptr<http::ClientConn> c) {
+    if (!c || !c->connected()) return;  // drop broken conns
+    std::lock_guard<std::mutex> lk(mu_);
+    auto& v = pool_[c->host() + ":" + std::to_string(c->port())];
+    if (v.size() < 64) v.push_back(std::move(c));
+  }
+
+ private:
+  std::mutex mu_;
+  std::map<std::string, std::vector<std::unique_ptr<http::ClientConn>>> pool_;
 };
 
 class GpuEngine {
@@ -253,7 +282,8 @@ class GpuEngine {
     for (int w = 0; w < num_conns; w++) {
       workers.emplace_back([&, w] {
         HIP_CHECK(hipSetDevice(device_));
-        http::ClientConn conn(u.host, u.port);
+        auto conn_holder = conn_pool_.checkout(u.host, u.port);
+        http::ClientConn& conn = *conn_holder;
         http::Headers h;
         for (auto& kv : headers) h[kv.first] = kv.second;
         while (true) {
@@ -313,6 +343,7 @@ class GpuEngine {
           }
           cv_pending_.notify_all();
         }
+        conn_pool_.checkin(std::move(conn_holder));
       });
     }
     for (auto& t : workers) t.join();
@@ -407,7 +438,8 @@ class GpuEngine {
     HIP_CHECK(hipSetDevice(device_));
     double t0 = now_s();
     http::Url u = http::Url::parse(url);
-    http::ClientConn conn(u.host, u.port);
+    auto conn_holder = conn_pool_.checkout(u.host, u.port);
+    http::ClientConn& conn = *conn_holder;
     http::Headers h;
     for (auto& kv : headers) h[kv.first] = kv.second;
     if (!conn.send_request(method, u.target(), h, static_cast<int64_t>(size)))
@@ -464,6 +496,7 @@ class GpuEngine {
     }
     release_slot(cur);
     release_slot(nxt);
+    conn_pool_.checkin(std::move(conn_holder));
     double t1 = now_s();
     if (status < 200 || status >= 300)
       throw std::runtime_error("push: HTTP " + std::to_string(status));
@@ -925,6 +958,7 @@ class GpuEngine {
   std::mutex zstd_mu_;
   void* zs_ptr_[7] = {};
   size_t zs_size_[7] = {};
+  ConnPool conn_pool_;
   std::mutex dedup_mu_;
   void* dedup_table_ = nullptr;
   uint64_t dedup_cap_ = 0;
