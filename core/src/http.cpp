@@ -446,6 +446,9 @@ void Server::accept_loop() {
 void Server::serve_conn(int fd, std::string peer) {
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+  int buf = 8 << 20;
+  setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof buf);
+  setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof buf);
   Conn conn(fd);
   if (ssl_ctx_) {
     SSL* ssl = SSL_new(static_cast<SSL_CTX*>(ssl_ctx_));
@@ -537,6 +540,11 @@ bool ClientConn::ensure_connected() {
   if (fd < 0) return false;
   int one = 1;
   setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+  // large buffers: the data plane moves multi-GiB objects over few
+  // sockets; default loopback buffers throttle per-conn throughput
+  int buf = 8 << 20;
+  setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &buf, sizeof buf);
+  setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &buf, sizeof buf);
   fd_ = fd;
   rbuf_.clear();
   rpos_ = 0;
