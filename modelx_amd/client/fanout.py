@@ -126,14 +126,23 @@ def fanout_pull_sharded(dist, gpu_client, repository: str, version: str, device:
     descs = [d for d in manifest.blobs if d.size > 0]
     plan = ShardPlan.build(descs, world)
     out = {}
-    for desc in descs:
-        owner = plan.owners[desc.name]
-        if owner == rank:
-            t = gpu_client.pull_blob_to_device(repository, desc, verify=verify)
-            out[desc.name] = t
-        elif replicate:
-            out[desc.name] = torch.empty(desc.size, dtype=torch.uint8,
-                                         device=f"cuda:{device}")
+    owned = [d for d in descs if plan.owners[d.name] == rank]
+    if owned:
+        # concurrent shard pulls (reentrant engine; per-blob latency would
+        # otherwise stack across a rank's whole 1/N of the index)
+        from concurrent.futures import ThreadPoolExecutor
+
+        with ThreadPoolExecutor(max_workers=min(len(owned), 6)) as ex:
+            for name, t in ex.map(
+                    lambda d: (d.name,
+                               gpu_client.pull_blob_to_device(repository, d, verify=verify)),
+                    owned):
+                out[name] = t
+    if replicate:
+        for desc in descs:
+            if plan.owners[desc.name] != rank:
+                out[desc.name] = torch.empty(desc.size, dtype=torch.uint8,
+                                             device=f"cuda:{device}")
     if replicate:
         # owners stream their blobs to everyone (binomial broadcast per blob;
         # different owners' broadcasts use disjoint xGMI links)
