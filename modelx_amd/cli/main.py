@@ -152,9 +152,19 @@ def cmd_version(args) -> int:
 def cmd_repo(args) -> int:
     mgr = RepoManager()
     if args.repo_cmd == "add":
-        token = args.token
-        mgr.set(RepoDetails(name=args.name, url=args.url, token=token or ""))
-        print(f"added repo {args.name} -> {args.url}")
+        # a ?token= in the URL moves into the stored token field
+        # (reference: ?token= URI auth, cmd/modelx/model/reference.go:61-63)
+        from urllib.parse import parse_qs
+
+        url, token = args.url, args.token
+        base, _, query = url.partition("?")
+        if query:
+            qtok = (parse_qs(query).get("token") or [""])[0]
+            if qtok and not token:
+                token = qtok
+                url = base
+        mgr.set(RepoDetails(name=args.name, url=url, token=token or ""))
+        print(f"added repo {args.name} -> {url}")
     elif args.repo_cmd == "remove":
         if not mgr.remove(args.name):
             print(f"repo {args.name} not found", file=sys.stderr)

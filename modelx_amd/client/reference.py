@@ -49,10 +49,12 @@ def parse_reference(raw: str, repo_manager: RepoManager = None) -> Reference:
         details = mgr.get(splits[0])  # raises if alias unknown
         if not auth and details.token:
             auth = "Bearer " + details.token
-        if len(splits) == 2:
-            raw = details.url + "/" + splits[1]
-        else:
-            raw = details.url
+        # the stored URL may carry its own query (?token=...): splice the
+        # repository path in BEFORE the query, not after it
+        base, _, query = details.url.partition("?")
+        raw = base.rstrip("/") + ("/" + splits[1] if len(splits) == 2 else "")
+        if query:
+            raw += "?" + query
 
     if not raw.startswith(("http://", "https://")):
         raw = "https://" + raw

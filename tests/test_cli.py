@@ -106,3 +106,42 @@ def test_modelxdl_filters_by_modelfiles(server, tmp_path, home):
     assert dl_main([f"{server.url}/proj/dl@v1", str(dest)]) == 0
     assert (dest / "wanted.bin").exists()
     assert not (dest / "ignored.bin").exists()
+
+
+class TestTokenAliasFlow:
+    """End-to-end CLI against an auth-enabled server through a repo alias
+    whose URL carried ?token= (reference.go:61-63 URI auth). Regression:
+    the alias query used to be spliced AFTER the path, mangling both the
+    repository path and the token."""
+
+    def test_full_flow(self, tmp_path, monkeypatch, capsys):
+        from util_servers import MODELXD, ServerProc, _build_servers, free_port, wait_http
+
+        from modelx_amd.cli.main import main as cli
+
+        _build_servers()
+        monkeypatch.setenv("HOME", str(tmp_path))
+        monkeypatch.delenv("MODELX_AUTH", raising=False)
+        port = free_port()
+        srv = ServerProc([MODELXD, "--listen", f"127.0.0.1:{port}", "--local-data",
+                          str(tmp_path / "data"), "--auth-tokens", "tok123"], port)
+        try:
+            wait_http(port)
+            assert cli(["repo", "add", "local", f"http://127.0.0.1:{port}?token=tok123"]) == 0
+            d = tmp_path / "model"
+            d.mkdir()
+            assert cli(["init", str(d)]) == 0
+            (d / "w.bin").write_bytes(os.urandom(64 * 1024))
+            assert cli(["push", "local/proj/demo@v1", str(d)]) == 0
+            assert cli(["list", "local"]) == 0
+            assert "proj/demo" in capsys.readouterr().out
+            assert cli(["list", "local/proj/demo@v1"]) == 0
+            assert "w.bin" in capsys.readouterr().out
+            out = tmp_path / "out"
+            assert cli(["pull", "local/proj/demo@v1", str(out)]) == 0
+            assert (out / "w.bin").read_bytes() == (d / "w.bin").read_bytes()
+            # wrong token is rejected
+            monkeypatch.setenv("MODELX_AUTH", "Bearer nope")
+            assert cli(["list", f"http://127.0.0.1:{port}/proj/demo"]) != 0
+        finally:
+            srv.stop()
