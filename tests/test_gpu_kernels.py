@@ -299,3 +299,52 @@ class TestChunkDedup:
         finally:
             mdx.stop()
             s3d.stop()
+
+
+class TestZstdCrossPath:
+    """GPU-pushed +zstd blobs pulled by the CPU client and vice versa —
+    the format is one (seekable multi-frame zstd), whichever side codes."""
+
+    def test_gpu_push_cpu_pull(self, tmp_path):
+        from util_servers import start_modelxd_s3, start_s3d
+
+        from modelx_amd.client import Client
+        from modelx_amd.client.gpu import GpuClient
+
+        s3d = start_s3d(str(tmp_path / "s3"))
+        mdx = start_modelxd_s3(s3d.url, redirect=True)
+        try:
+            g = GpuClient(mdx.url, device=0, num_slots=4, slot_bytes=8 << 20)
+            data = (b"interop " * 8192 + os.urandom(1024)) * 64  # ~4 MiB
+            src = torch.frombuffer(bytearray(data), dtype=torch.uint8).cuda()
+            g.push_from_gpu("xp/gpu2cpu", "v1", {"weights.bin": src}, compress="zstd")
+            c = Client(mdx.url)
+            out = tmp_path / "out"
+            c.pull("xp/gpu2cpu", "v1", str(out), quiet=True)
+            assert (out / "weights.bin").read_bytes() == data
+        finally:
+            mdx.stop()
+            s3d.stop()
+
+    def test_cpu_push_gpu_pull(self, tmp_path):
+        from util_servers import start_modelxd_s3, start_s3d
+
+        from modelx_amd.client import Client
+        from modelx_amd.client.gpu import GpuClient
+        from modelx_amd.config import ModelConfig
+
+        s3d = start_s3d(str(tmp_path / "s3"))
+        mdx = start_modelxd_s3(s3d.url, redirect=True)
+        try:
+            d = tmp_path / "model"
+            d.mkdir()
+            (d / "modelx.yaml").write_text(ModelConfig(description="xp").to_yaml())
+            data = (b"cpu-coded " * 6553 + os.urandom(512)) * 48  # ~3 MiB
+            (d / "weights.bin").write_bytes(data)
+            Client(mdx.url).push("xp/cpu2gpu", "v1", str(d), quiet=True, compress="zstd")
+            g = GpuClient(mdx.url, device=0, num_slots=4, slot_bytes=8 << 20)
+            out = g.pull_to_gpu("xp/cpu2gpu", "v1")
+            assert bytes(out["weights.bin"].cpu().numpy().tobytes()) == data
+        finally:
+            mdx.stop()
+            s3d.stop()
