@@ -122,16 +122,21 @@ def cmd_login(args) -> int:
     """Verify access then store the token (reference: login.go:51-62)."""
     ref, _ = _client_for(args.ref, args.insecure)
     token = args.token
-    if not token and sys.stdin.isatty():
-        import getpass
+    if not token:
+        if sys.stdin.isatty():
+            import getpass
 
-        token = getpass.getpass("token: ")
+            token = getpass.getpass("token: ")
+        else:  # piped: modelx login NAME <<< "$TOKEN"
+            token = sys.stdin.readline().strip()
     from ..client import Client
 
     client = Client(ref.registry, f"Bearer {token}" if token else "", insecure=args.insecure)
     client.ping()
     mgr = RepoManager()
-    name = args.name or ref.registry.split("//", 1)[-1].split(":")[0]
+    default_name = args.ref if "://" not in args.ref else \
+        ref.registry.split("//", 1)[-1].split(":")[0]
+    name = args.name or default_name.split("/", 1)[0]
     mgr.set(RepoDetails(name=name, url=ref.registry, token=token or ""))
     print(f"login succeeded: {name} -> {ref.registry}")
     return 0
