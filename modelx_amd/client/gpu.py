@@ -9,8 +9,9 @@ core/src/engine.cpp + core/hip/sha256.hip):
   torch.Tensor]`` (uint8) resident in HBM.
 - ``push_from_gpu``: chunk-digest the device buffers on GPU, presigned
   (multi)part upload streamed D2H through the pinned ring, manifest PUT last.
-- multi-GPU fan-out (``broadcast_pull``): rank 0 pulls, RCCL broadcast over
-  xGMI per pipeline chunk via torch.distributed — see ``fanout.py``.
+- multi-GPU fan-out: rank 0 pulls, RCCL broadcast over xGMI per pipeline
+  chunk via torch.distributed — see ``fanout.py`` (fanout_pull_broadcast /
+  fanout_pull_sharded).
 
 The native extension is REQUIRED on a GPU box: if HIP devices are visible and
 the extension is missing, this module raises instead of silently falling back
@@ -33,7 +34,7 @@ DEFAULT_PART_BYTES = 256 << 20  # push part granularity (multipart)
 DEFAULT_PUSH_PARALLEL = 6
 # 128 KiB chunks put 65k+ SHA-256 chains in flight for multi-GiB blobs — the
 # measured sweet spot on MI355X (1017 GiB/s vs 147 GiB/s at 1 MiB chunks,
-# profiles/gpu_micro.md)
+# profiles/sha256_kernel.md)
 DEFAULT_GPU_CHUNK = 128 << 10
 
 
