@@ -123,7 +123,10 @@ def config5(args, g, dist, rank, world, device):
     small_sz = 64 << 20
     huge_n = 2
     huge_sz = max(int(120 * args.scale * (1 << 30)), 64 << 20)
-    page = torch.randint(0, 256, (1 << 20,), dtype=torch.uint8, device=f"cuda:{device}")
+    # 64 KiB tile period: repeats land WITHIN each 128 KiB zstd frame (frames
+    # are independent, so a period larger than the frame defeats compression)
+    # and every 128 KiB dedup chunk of the tiled region is identical
+    page = torch.randint(0, 256, (64 << 10,), dtype=torch.uint8, device=f"cuda:{device}")
 
     def make(nbytes: int, seed: int) -> torch.Tensor:
         # tiled shared pages (dedup + compressible) + unique head
