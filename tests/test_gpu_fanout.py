@@ -189,12 +189,13 @@ def test_modelxdl_gpus_flag(stack, tmp_path, capsys):
     from modelx_amd.client.gpu import GpuClient
     from modelx_amd.config import ModelConfig
 
-    g = GpuClient(stack.url, device=0, num_slots=4, slot_bytes=8 << 20)
+    mdx, _ = stack
+    g = GpuClient(mdx.url, device=0, num_slots=4, slot_bytes=8 << 20)
     w = torch.randint(0, 256, (8 << 20,), dtype=torch.uint8, device="cuda:0")
     cfg = ModelConfig(description="dl", model_files=["weights.bin"])
     g.push_from_gpu("dl/model", "v1", {"weights.bin": w, "skipme.bin": w[: 1 << 20]},
                     config_yaml=cfg.to_yaml())
-    rc = dl_main([f"{stack.url}/dl/model@v1", "--gpus", "0"])
+    rc = dl_main([f"{mdx.url}/dl/model@v1", "--gpus", "0"])
     assert rc == 0
     out = capsys.readouterr().out
     assert "weights.bin" in out and "skipme.bin" not in out
