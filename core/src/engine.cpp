@@ -240,12 +240,18 @@ class GpuEngine {
     // per-call pull state: pull_impl is reentrant — concurrent pulls (e.g.
     // many small blobs of one index from a Python thread pool) share the
     // pinned-slot pool and streams but own their range queue and error.
+    // one HTTP request covers several pinned slots ("super-range"): the
+    // response streams into consecutive slots, quartering per-request
+    // overhead (headers, server sendfile setup) on big pulls. Explicit
+    // range lists (dedup/refetch) are used as given.
+    constexpr uint64_t kSlotsPerRequest = 4;
     std::vector<Range> ranges;
     if (explicit_ranges) {
       ranges = *explicit_ranges;
     } else {
-      for (uint64_t off = 0; off < size; off += slot_bytes_)
-        ranges.push_back({off, std::min<uint64_t>(slot_bytes_, size - off)});
+      uint64_t super = slot_bytes_ * kSlotsPerRequest;
+      for (uint64_t off = 0; off < size; off += super)
+        ranges.push_back({off, std::min<uint64_t>(super, size - off)});
     }
     std::atomic<size_t> next_range{0};
     std::mutex err_mu;
@@ -297,51 +303,64 @@ class GpuEngine {
             }
             r = ranges[i];
           }
-          Slot* slot = acquire_slot();
+          // fetch the whole super-range in one request, landing each
+          // slot-sized piece as it streams in; on failure retry the whole
+          // super-range on a fresh connection (pieces are idempotent)
           bool ok = false;
           for (int attempt = 0; attempt < 3 && !ok; attempt++) {
             double tn = now_s();
-            ok = fetch_range(conn, u, h, {r.offset + base_offset, r.length}, slot->host);
+            if (!begin_range_fetch(conn, u, h, {r.offset + base_offset, r.length})) continue;
+            ok = true;
+            for (uint64_t poff = 0; poff < r.length && ok; poff += slot_bytes_) {
+              uint64_t plen = std::min<uint64_t>(slot_bytes_, r.length - poff);
+              Slot* slot = acquire_slot();
+              if (!conn.read_body_exact(slot->host, plen)) {
+                conn.close_fd();
+                release_slot(slot);
+                ok = false;
+                break;
+              }
+              uint64_t abs_off = r.offset + poff;
+              hipStream_t st = streams_[stream_rr.fetch_add(1) % streams_.size()];
+              hipError_t e = hipMemcpyAsync(reinterpret_cast<char*>(dst_ptr) + abs_off,
+                                            slot->host, plen, hipMemcpyHostToDevice, st);
+              if (e == hipSuccess && hash_chunk) {
+                uint32_t first_chunk = static_cast<uint32_t>(abs_off / hash_chunk);
+                uint32_t n_chunks =
+                    static_cast<uint32_t>((plen + hash_chunk - 1) / hash_chunk);
+                e = modelx_sha256_chunk_leaves(
+                    reinterpret_cast<char*>(dst_ptr) + abs_off, plen, hash_chunk,
+                    static_cast<char*>(dleaves) + static_cast<size_t>(first_chunk) * 32,
+                    n_chunks, st);
+              }
+              if (e == hipSuccess) e = hipEventRecord(slot->event, st);
+              if (e != hipSuccess) {
+                {
+                  std::lock_guard<std::mutex> lk(err_mu);
+                  if (error.empty()) error = std::string("hip: ") + hipGetErrorString(e);
+                }
+                release_slot(slot);
+                ok = false;
+                // HIP errors are not retryable
+                attempt = 3;
+                break;
+              }
+              {
+                std::lock_guard<std::mutex> lk(mu_);
+                pending_.push_back(slot);
+              }
+              cv_pending_.notify_all();
+            }
             if (ok) {
               net_ns.fetch_add(static_cast<long>((now_s() - tn) * 1e9));
               net_bytes.fetch_add(r.length);
             }
           }
           if (!ok) {
-            {
-              std::lock_guard<std::mutex> lk(err_mu);
-              if (error.empty()) error = "range fetch failed @" + std::to_string(r.offset);
-            }
-            release_slot(slot);
+            std::lock_guard<std::mutex> lk(err_mu);
+            if (error.empty()) error = "range fetch failed @" + std::to_string(r.offset);
             break;
           }
-          hipStream_t st = streams_[stream_rr.fetch_add(1) % streams_.size()];
-          hipError_t e = hipMemcpyAsync(reinterpret_cast<char*>(dst_ptr) + r.offset, slot->host,
-                                        r.length, hipMemcpyHostToDevice, st);
-          if (e == hipSuccess && hash_chunk) {
-            // hash this slot's chunks right behind its copy (stream-ordered)
-            uint32_t first_chunk = static_cast<uint32_t>(r.offset / hash_chunk);
-            uint32_t n_chunks = static_cast<uint32_t>((r.length + hash_chunk - 1) / hash_chunk);
-            uint64_t span = std::min<uint64_t>(r.length, size - r.offset);
-            e = modelx_sha256_chunk_leaves(
-                reinterpret_cast<char*>(dst_ptr) + r.offset, span, hash_chunk,
-                static_cast<char*>(dleaves) + static_cast<size_t>(first_chunk) * 32, n_chunks,
-                st);
-          }
-          if (e == hipSuccess) e = hipEventRecord(slot->event, st);
-          if (e != hipSuccess) {
-            {
-              std::lock_guard<std::mutex> lk(err_mu);
-              if (error.empty()) error = std::string("hip: ") + hipGetErrorString(e);
-            }
-            release_slot(slot);
-            break;
-          }
-          {
-            std::lock_guard<std::mutex> lk(mu_);
-            pending_.push_back(slot);
-          }
-          cv_pending_.notify_all();
         }
         conn_pool_.checkin(std::move(conn_holder));
       });
@@ -918,27 +937,21 @@ class GpuEngine {
     cv_free_.notify_all();
   }
 
-  // ranged GET into `dst` (pinned); returns false on any protocol error
-  bool fetch_range(http::ClientConn& conn, const http::Url& u, http::Headers h, const Range& r,
-                   char* dst) {
+  // start a ranged GET and validate the response head; body is then read
+  // with read_body_exact by the caller. STRICT framing: must answer 206
+  // with Content-Length == the requested span — a 200 (full object) or a
+  // mismatched length would land the wrong bytes silently.
+  bool begin_range_fetch(http::ClientConn& conn, const http::Url& u, http::Headers h,
+                         const Range& r) {
     h["Range"] =
         "bytes=" + std::to_string(r.offset) + "-" + std::to_string(r.offset + r.length - 1);
     if (!conn.send_request("GET", u.target(), h, -1)) return false;
     int status = 0;
     http::Headers rh;
     if (!conn.read_response_head(&status, &rh)) return false;
-    // STRICT framing: a ranged GET must answer 206 with Content-Length ==
-    // the requested span. A 200 (full object) or a mismatched length would
-    // land the wrong bytes silently; fail the fetch and drop the
-    // connection so the retry starts clean.
     auto cl = rh.find("Content-Length");
     if (status != 206 || cl == rh.end() ||
         atoll(cl->second.c_str()) != static_cast<long long>(r.length)) {
-      conn.close_fd();
-      return false;
-    }
-    // bulk MSG_WAITALL read straight into the pinned slot
-    if (!conn.read_body_exact(dst, r.length)) {
       conn.close_fd();
       return false;
     }
