@@ -211,8 +211,30 @@ class GpuClient:
                                  f"+zstd blob {desc.name} lacks the raw-size annotation")
         out = torch.empty(max(raw_size, 1), dtype=torch.uint8, device=f"cuda:{self.device}")
         t0 = time.monotonic()
-        got = self.engine.zstd_decompress_device(comp.data_ptr(), desc.size,
-                                                 out.data_ptr(), out.numel())
+        try:
+            got = self.engine.zstd_decompress_device(comp.data_ptr(), desc.size,
+                                                     out.data_ptr(), out.numel())
+        except RuntimeError as e:
+            # diagnostic: persist the (digest-verified) compressed bytes so a
+            # decode failure is analyzable offline; then retry once — a
+            # second identical failure is deterministic (bad stored frame),
+            # a pass means a device-side race to hunt
+            import sys
+
+            try:
+                os_dir = "gpurun_out"
+                import os as _os
+
+                _os.makedirs(os_dir, exist_ok=True)
+                path = f"{os_dir}/zstd_decode_fail_{desc.digest.split(':')[-1][:12]}.zst"
+                with open(path, "wb") as f:
+                    f.write(bytes(comp[: desc.size].cpu().numpy().tobytes()))
+                print(f"modelx: zstd decode failed ({e}); dumped {path}; retrying once",
+                      file=sys.stderr)
+            except Exception:
+                pass
+            got = self.engine.zstd_decompress_device(comp.data_ptr(), desc.size,
+                                                     out.data_ptr(), out.numel())
         self.last_stats.append({"phase": "pull-zstd-decompress", "bytes": got,
                                 "seconds": time.monotonic() - t0})
         if raw_size and got != raw_size:
