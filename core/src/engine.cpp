@@ -431,8 +431,14 @@ class GpuEngine {
     std::string out;
     {
       py::gil_scoped_release release;
-      HIP_CHECK(hipMemcpy(dptrs, hptrs.data(), n * sizeof(void*), hipMemcpyHostToDevice));
-      HIP_CHECK(hipMemcpy(dlens, hlens.data(), n * sizeof(uint64_t), hipMemcpyHostToDevice));
+      // stream-ordered staging: a plain hipMemcpy is NOT ordered against a
+      // kernel launched on a non-blocking stream — the kernel could read
+      // stale parameters (observed as rare packed-frame corruption in the
+      // equivalent zstd path)
+      HIP_CHECK(hipMemcpyAsync(dptrs, hptrs.data(), n * sizeof(void*),
+                               hipMemcpyHostToDevice, hash_stream_));
+      HIP_CHECK(hipMemcpyAsync(dlens, hlens.data(), n * sizeof(uint64_t),
+                               hipMemcpyHostToDevice, hash_stream_));
       HIP_CHECK(modelx_sha256_multibuf(reinterpret_cast<const void* const*>(dptrs),
                                        reinterpret_cast<const uint64_t*>(dlens), n, ddig,
                                        hash_stream_));
@@ -572,8 +578,8 @@ class GpuEngine {
         segs[i] = {entries[i].header_off, reinterpret_cast<uint64_t>(dgather) + i * 512, 512};
       void* dsegs = nullptr;
       HIP_CHECK(hipMalloc(&dsegs, segs.size() * sizeof(CopySegHost)));
-      HIP_CHECK(hipMemcpy(dsegs, segs.data(), segs.size() * sizeof(CopySegHost),
-                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpyAsync(dsegs, segs.data(), segs.size() * sizeof(CopySegHost),
+                               hipMemcpyHostToDevice, hash_stream_));
       HIP_CHECK(modelx_tar_scatter(reinterpret_cast<void*>(tar_ptr), dsegs, count,
                                    hash_stream_));
       HIP_CHECK(hipStreamSynchronize(hash_stream_));
@@ -656,8 +662,8 @@ class GpuEngine {
     }
     void* dsegs = nullptr;
     HIP_CHECK(hipMalloc(&dsegs, pieces.size() * sizeof(CopySegHost)));
-    HIP_CHECK(hipMemcpy(dsegs, pieces.data(), pieces.size() * sizeof(CopySegHost),
-                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpyAsync(dsegs, pieces.data(), pieces.size() * sizeof(CopySegHost),
+                             hipMemcpyHostToDevice, hash_stream_));
     HIP_CHECK(modelx_tar_scatter(reinterpret_cast<void*>(tar_ptr), dsegs,
                                  static_cast<uint32_t>(pieces.size()), hash_stream_));
     HIP_CHECK(hipStreamSynchronize(hash_stream_));
@@ -736,14 +742,16 @@ class GpuEngine {
         out += (uint64_t)hsizes[i];
         if (out > dst_cap) throw std::runtime_error("zstd compress: dst overflow");
       }
-      HIP_CHECK(hipMemcpy(dsegs, hsegs.data(), n * sizeof(CopySegHost), hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpyAsync(dsegs, hsegs.data(), n * sizeof(CopySegHost),
+                               hipMemcpyHostToDevice, hash_stream_));
       HIP_CHECK(modelx_tar_scatter(dscratch, dsegs, n, hash_stream_));
       HIP_CHECK(hipStreamSynchronize(hash_stream_));
     }
     auto table = zstdhost::build_seek_table(entries);
     if (out + table.size() > dst_cap) throw std::runtime_error("zstd compress: dst overflow");
-    HIP_CHECK(hipMemcpy(reinterpret_cast<char*>(dst_ptr) + out, table.data(), table.size(),
-                        hipMemcpyHostToDevice));
+    HIP_CHECK(hipMemcpyAsync(reinterpret_cast<char*>(dst_ptr) + out, table.data(),
+                             table.size(), hipMemcpyHostToDevice, hash_stream_));
+    HIP_CHECK(hipStreamSynchronize(hash_stream_));
     out += table.size();
     return out;
   }
@@ -796,8 +804,8 @@ class GpuEngine {
     std::vector<int64_t> hrc(batch);
     for (uint64_t first = 0; first < nframes; first += batch) {
       uint32_t n = (uint32_t)std::min<uint64_t>(batch, nframes - first);
-      HIP_CHECK(hipMemcpy(dframes, frames.data() + first, n * sizeof(MxzFrameHost),
-                          hipMemcpyHostToDevice));
+      HIP_CHECK(hipMemcpyAsync(dframes, frames.data() + first, n * sizeof(MxzFrameHost),
+                               hipMemcpyHostToDevice, hash_stream_));
       HIP_CHECK(modelx_zstd_decompress_frames(reinterpret_cast<void*>(src_ptr), dframes, n,
                                               reinterpret_cast<void*>(dst_ptr), dlit, drc,
                                               hash_stream_));
