@@ -247,6 +247,13 @@ struct BuildScratch {
 // Build a decoding table from normalized counts (-1 == "less than 1").
 MX_HD static inline int fse_build_dtable(FseTable* t, const i16* counts, u32 nsym, u32 log,
                                          BuildScratch* bs) {
+  // Barrier before overwriting shared scratch/tables: __syncthreads only
+  // bounds wave drift to ONE barrier interval, and a whole block's table
+  // builds share an interval — without this, a leading wave overwrites
+  // bs->spread / the target table while a trailing wave still reads the
+  // previous section's content (observed ~1/50k frames as rc=-7 or wrong
+  // output under multi-wave decode).
+  mx_sync();
   if (log > kMaxFseLog) return MXZ_ERR_FSE;
   u32 size = 1u << log;
   u32 high = size - 1;
@@ -387,6 +394,7 @@ MX_HD static inline int huf_build(HufTable* t, const u8* weights, u32 nsym, u32 
 // Parse a Huffman_Tree_Description; returns bytes consumed or <0.
 MX_HD static inline int huf_read_table(HufTable* t, const u8* src, u64 srclen,
                                        BuildScratch* bs) {
+  mx_sync();  // see fse_build_dtable: shared-scratch overwrite boundary
   if (srclen < 1) return MXZ_ERR_HUFFMAN;
   u8* weights = bs->weights;
   u32 nsym = 0;
@@ -554,6 +562,7 @@ MX_HD static inline void dec_ctx_init(DecCtx* c, u8* lit_scratch) {
 // consumed or <0.  mode: 0 predefined, 1 RLE, 2 FSE, 3 repeat.
 MX_HD static inline int seq_table_load(FseTable* t, bool* valid, u32 mode, u32 kind,
                                        const u8* src, u64 srclen, BuildScratch* bs) {
+  mx_sync();  // see fse_build_dtable: shared-scratch overwrite boundary
   // kind: 0=LL 1=OF 2=ML
   if (mode == 0) {
     u32 nsym, log;
@@ -597,6 +606,10 @@ MX_HD static inline int seq_table_load(FseTable* t, bool* valid, u32 mode, u32 k
 // the frame buffer before dst). Returns regenerated size or <0.
 MX_HD static inline i64 decode_block(DecCtx* c, u32* rep, const u8* src, u64 srclen, u8* dst,
                                      u64 dstcap, u64 history) {
+  // block boundary: the previous block's sequence loop read lit_scratch and
+  // the FSE/huffman tables; make every wave finish those reads before this
+  // block overwrites them (see fse_build_dtable note)
+  mx_sync();
   // ---- literals section ----
   if (srclen < 1) return MXZ_ERR_LITERALS;
   u32 b0 = src[0];
