@@ -127,3 +127,40 @@ def test_size_mismatch_rejected_at_manifest_put(stack, tmp_path):
         c.remote.put_manifest("proj/szbad", "v1", manifest)
     # mismatched blob was deleted server-side
     assert not c.remote.head_blob("proj/szbad", blob.digest)
+
+
+def test_presigned_url_expiry_enforced(tmp_path):
+    """Expired presigned URLs are rejected by the object server
+    (reference: --s3-presign-expire, pkg/registry/fs_s3.go:37; sigv4
+    expiry check is ours — MinIO does the same)."""
+    import time
+
+    import requests
+
+    from util_servers import MODELXD, ServerProc, free_port, start_s3d, wait_http
+
+    s3d = start_s3d(str(tmp_path / "s3"))
+    port = free_port()
+    mdx = ServerProc([MODELXD, "--listen", f"127.0.0.1:{port}", "--s3-url", s3d.url,
+                      "--s3-bucket", "modelx", "--s3-access-key", "modelx",
+                      "--s3-secret-key", "modelx123", "--enable-redirect",
+                      "--s3-presign-expire", "1"], port)
+    try:
+        wait_http(port)
+        d = tmp_path / "m"
+        d.mkdir()
+        (d / "modelx.yaml").write_text(ModelConfig(description="exp").to_yaml())
+        (d / "w.bin").write_bytes(os.urandom(256 * 1024))
+        c = Client(mdx.url)
+        c.push("exp/model", "v1", str(d), quiet=True)
+        manifest = c.get_manifest("exp/model", "v1")
+        (desc,) = [b for b in manifest.blobs if b.name == "w.bin"]
+        loc = c.remote.get_blob_location("exp/model", desc, "download")
+        url = loc.properties["parts"][0]["url"]
+        assert requests.get(url, timeout=10).status_code == 200  # fresh: works
+        time.sleep(2.5)
+        r = requests.get(url, timeout=10)
+        assert r.status_code == 403, f"expired presign must 403, got {r.status_code}"
+    finally:
+        mdx.stop()
+        s3d.stop()
