@@ -292,13 +292,216 @@ MX_HD static inline u32 lz_parse(const u8* base, u64 block_off, u64 block_len, u
 }
 #endif
 
+// ------------------------------------------------- huffman literals --------
+
+// Byte histogram. Device: lane-strided with LDS atomics (single-wave
+// kernel; all lanes see the final counts after the barrier).
+MX_HD static inline void lit_histogram(const u8* lit, u64 n, u32* hist) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  u32 lane = mx_lane(), w = mx_width();
+  for (u32 i = lane; i < 256; i += w) hist[i] = 0;
+  mx_sync();
+  for (u64 i = lane; i < n; i += w) atomicAdd(&hist[lit[i]], 1u);
+  mx_sync();
+#else
+  for (u32 i = 0; i < 256; i++) hist[i] = 0;
+  for (u64 i = 0; i < n; i++) hist[lit[i]]++;
+#endif
+}
+
+struct HufEnc {
+  u16 code[256];
+  u8 len[256];  // 0 = absent
+  u32 maxbits;
+  u32 last_sym;
+  u64 est_bits;
+};
+
+// Kraft-exact length-limited (<=11) code construction + canonical code
+// assignment matching the decoder's table fill (huf_build: weight
+// ascending == length descending, symbol ascending). Returns false when
+// huffman can't apply (degenerate alphabet, or symbols past 127 — the
+// direct 4-bit weight serialization covers <=128 explicit weights and we
+// fall back to raw literals beyond that).
+MX_HD static inline bool huf_build_enc(const u32* hist, u64 total, HufEnc* e) {
+  u32 nsym = 0, last = 0;
+  for (u32 s = 0; s < 256; s++)
+    if (hist[s]) {
+      nsym++;
+      last = s;
+    }
+  if (nsym < 2 || last > 127 || total < 64) return false;
+  for (u32 s = 0; s < 256; s++) e->len[s] = 0;
+  i64 K = 0;  // kraft sum in units of 2^-11
+  for (u32 s = 0; s <= last; s++) {
+    if (!hist[s]) continue;
+    u32 ratio = (u32)(total / hist[s]);
+    u32 l = ratio <= 1 ? 1 : mx_highbit(ratio) + ((ratio & (ratio - 1)) ? 1 : 0);
+    if (l < 1) l = 1;
+    if (l > 11) l = 11;
+    e->len[s] = (u8)l;
+    K += (i64)1 << (11 - l);
+  }
+  // shrink shares of the rarest symbols until the sum fits
+  for (int guard = 0; K > 2048 && guard < 4096; guard++) {
+    u32 pick = 256;
+    u32 best = 0xFFFFFFFFu;
+    for (u32 s = 0; s <= last; s++)
+      if (e->len[s] && e->len[s] < 11 && hist[s] < best) {
+        best = hist[s];
+        pick = s;
+      }
+    if (pick == 256) return false;
+    K -= (i64)1 << (11 - e->len[pick] - 1);
+    e->len[pick]++;
+  }
+  // grow shares of the most frequent symbols to land exactly on 2^11
+  for (int guard = 0; K < 2048 && guard < 4096; guard++) {
+    u32 pick = 256;
+    u32 best = 0;
+    i64 gap = 2048 - K;
+    for (u32 s = 0; s <= last; s++) {
+      if (!e->len[s] || e->len[s] <= 1) continue;
+      i64 delta = (i64)1 << (11 - e->len[s]);
+      if (delta <= gap && hist[s] >= best) {
+        best = hist[s];
+        pick = s;
+      }
+    }
+    if (pick == 256) return false;
+    K += (i64)1 << (11 - e->len[pick]);
+    e->len[pick]--;
+  }
+  if (K != 2048) return false;
+  u32 maxbits = 0;
+  for (u32 s = 0; s <= last; s++)
+    if (e->len[s] > maxbits) maxbits = e->len[s];
+  // canonical codes in the decoder's fill order
+  u32 size = 1u << maxbits;
+  u32 pos = 0;
+  for (u32 l = maxbits; l >= 1; l--) {
+    u32 run = 1u << (maxbits - l);
+    for (u32 s = 0; s <= last; s++) {
+      if (e->len[s] != l) continue;
+      e->code[s] = (u16)(pos >> (maxbits - l));
+      pos += run;
+    }
+  }
+  if (pos != size) return false;
+  u64 bits = 0;
+  for (u32 s = 0; s <= last; s++) bits += (u64)hist[s] * e->len[s];
+  e->maxbits = maxbits;
+  e->last_sym = last;
+  e->est_bits = bits;
+  return true;
+}
+
+// Encode lit[a..b) as one backward huffman stream; returns bytes or <0.
+MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a, u64 b,
+                                          u8* dst, u64 cap) {
+  BitW bw;
+  bw.init(dst, cap);
+  // written back-to-front so the backward reader produces them in order
+  for (i64 i = (i64)b - 1; i >= (i64)a; i--) {
+    u8 s = lit[i];
+    bw.add(e->code[s], e->len[s]);
+    if (bw.overflow) return MXZ_ERR_DST_SMALL;
+  }
+  bw.close();
+  if (bw.overflow) return MXZ_ERR_DST_SMALL;
+  return (i64)bw.pos;
+}
+
+// Emit a Compressed_Literals_Block (4-stream huffman, direct 4-bit weight
+// table). Returns total section bytes (header included) or <0 when raw is
+// better / capacity exceeded.
+MX_HD static inline i64 emit_huf_literals(const HufEnc* e, const u8* lit, u64 n, u8* out,
+                                          u64 cap) {
+  if (n < 256) return MXZ_ERR_DST_SMALL;  // not worth the table
+  // table: header byte + packed 4-bit weights for symbols 0..last_sym-1
+  u32 nweights = e->last_sym;  // last symbol's weight is implied
+  u64 tbl = 1 + (nweights + 1) / 2;
+  u64 est_total = 5 + tbl + 6 + (e->est_bits + 7) / 8 + 8;
+  if (est_total >= n) return MXZ_ERR_DST_SMALL;
+  // header needs the compressed size — assemble body first at a safe
+  // offset (max header 5 bytes), then write the header knowing sizes
+  u64 hmax = 5;
+  if (hmax + tbl + 6 >= cap) return MXZ_ERR_DST_SMALL;
+  u8* body = out + hmax;
+  u64 bcap = cap - hmax;
+  body[0] = (u8)(127 + nweights);
+  for (u32 i = 0; i < nweights; i++) {
+    u32 l = e->len[i];
+    u32 wgt = l ? (e->maxbits + 1 - l) : 0;
+    if (i & 1)
+      body[1 + i / 2] |= (u8)wgt;
+    else
+      body[1 + i / 2] = (u8)(wgt << 4);
+  }
+  u64 bpos = tbl;
+  u8* jump = body + bpos;
+  bpos += 6;
+  u64 r123 = (n + 3) / 4;
+  u64 r4 = n - 3 * r123;
+  if (r4 == 0) return MXZ_ERR_DST_SMALL;  // stream 4 must be non-empty
+  u64 sizes[4];
+  u64 offs[4] = {0, r123, 2 * r123, 3 * r123};
+  u64 lens[4] = {r123, r123, r123, r4};
+  for (int k = 0; k < 4; k++) {
+    i64 m = huf_encode_stream(e, lit, offs[k], offs[k] + lens[k], body + bpos, bcap - bpos);
+    if (m < 0) return m;
+    if (k < 3 && m > 0xFFFF) return MXZ_ERR_DST_SMALL;
+    sizes[k] = (u64)m;
+    bpos += (u64)m;
+    if (bpos + 64 > bcap) return MXZ_ERR_DST_SMALL;
+  }
+  jump[0] = (u8)sizes[0];
+  jump[1] = (u8)(sizes[0] >> 8);
+  jump[2] = (u8)sizes[1];
+  jump[3] = (u8)(sizes[1] >> 8);
+  jump[4] = (u8)sizes[2];
+  jump[5] = (u8)(sizes[2] >> 8);
+  u64 comp = bpos;  // table + jump + streams
+  if (comp >= n) return MXZ_ERR_DST_SMALL;
+  // literals header: type=2 (Compressed), 4 streams, size-format 11
+  // (18-bit fields — always large enough, and a fixed 5-byte header means
+  // the body was assembled at its final offset)
+  u64 h = 2u | (3u << 2);
+  h |= n << 4;
+  h |= comp << 22;
+  for (u64 i = 0; i < 5; i++) out[i] = (u8)(h >> (8 * i));
+  return (i64)(5 + comp);
+}
+
 // ----------------------------------------------------------- block emit ----
 
+// Predefined-table FSE encoders + literal histogram (caller-allocated:
+// LDS on the GPU so the redundant-wavefront execution doesn't spill
+// per-lane copies; the histogram is filled with LDS atomics).
+struct EncTables {
+  FseEnc ell, eof, eml;
+  HufEnc he;
+  u32 lit_hist[256];
+};
+
+MX_HD static inline int enc_tables_init(EncTables* et) {
+  u32 nsym, log;
+  const i16* d = ll_default_dist(&nsym, &log);
+  if (fse_build_ctable(&et->ell, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  d = of_default_dist(&nsym, &log);
+  if (fse_build_ctable(&et->eof, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  d = ml_default_dist(&nsym, &log);
+  if (fse_build_ctable(&et->eml, d, nsym, log) < 0) return MXZ_ERR_FSE;
+  return MXZ_OK;
+}
+
 // Emit one zstd block (compressed if it wins, raw otherwise) for
-// src[block_off..block_off+block_len). Returns bytes written or <0.
+// src[block_off..block_off+block_len). Literals pick the cheapest of
+// RLE / 4-stream huffman / raw; a whole-block single byte becomes an RLE
+// block. Returns bytes written or <0.
 MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len, bool last,
                                      u8* dst, u64 dstcap, u32* hash, Seq* seqs, u32 max_seqs,
-                                     const FseEnc* ell, const FseEnc* eof, const FseEnc* eml) {
+                                     EncTables* et) {
   const u8* block = src + block_off;
   u64 raw_total = 3 + block_len;
   if (dstcap < raw_total) return MXZ_ERR_DST_SMALL;
@@ -306,38 +509,107 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
   u64 lit_total = 0;
   u32 nseq = lz_parse(src, block_off, block_len, hash, seqs, max_seqs, &lit_total);
 
-  bool use_raw = nseq == 0;
   u64 csize = 0;
-  if (!use_raw) {
-    // literals header (raw literals)
+  bool use_raw = false;
+  if (nseq == 0) {
+    // no matches: RLE block for a single repeated byte, else try a
+    // huffman-literals-only compressed block, else raw
+    lit_histogram(block, block_len, et->lit_hist);
+    u32 nsym = 0;
+    for (u32 i = 0; i < 256; i++)
+      if (et->lit_hist[i]) nsym++;
+    if (nsym == 1 && block_len > 0 && block_len < (1u << 21)) {
+      u32 bh = ((u32)block_len << 3) | (1u << 1) | (last ? 1 : 0);
+      dst[0] = (u8)bh;
+      dst[1] = (u8)(bh >> 8);
+      dst[2] = (u8)(bh >> 16);
+      dst[3] = block[0];
+      return 4;
+    }
+    if (block_len >= 512 && huf_build_enc(et->lit_hist, block_len, &et->he)) {
+      i64 n = emit_huf_literals(&et->he, block, block_len, dst + 3, dstcap - 5);
+      if (n > 0 && (u64)n + 1 < block_len) {
+        dst[3 + n] = 0;  // zero sequences
+        csize = (u64)n + 1;
+        u32 bh = ((u32)csize << 3) | (2u << 1) | (last ? 1 : 0);
+        dst[0] = (u8)bh;
+        dst[1] = (u8)(bh >> 8);
+        dst[2] = (u8)(bh >> 16);
+        return (i64)(3 + csize);
+      }
+    }
+    use_raw = true;
+  }
+
+  if (!use_raw && nseq > 0) {
     u8* out = dst + 3;
     u64 cap = dstcap - 3;
-    u64 lp = 0;
-    if (lit_total <= 31) {
-      out[lp++] = (u8)(0 | (0 << 2) | (lit_total << 3));
-    } else if (lit_total <= 4095) {
-      out[lp++] = (u8)(0 | (1 << 2) | ((lit_total & 0xF) << 4));
-      out[lp++] = (u8)(lit_total >> 4);
-    } else {
-      out[lp++] = (u8)(0 | (3 << 2) | ((lit_total & 0xF) << 4));
-      out[lp++] = (u8)((lit_total >> 4) & 0xFF);
-      out[lp++] = (u8)(lit_total >> 12);
-    }
-    if (lp + lit_total + 16 > cap) {
-      use_raw = true;
-    } else {
-      // literal bytes: replay the parse to copy each run
+    // assemble the literal bytes into the tail of the seqs scratch
+    // (capacity proof: lit_total <= block_len - 4*nseq and the free tail
+    // is 12*(max_seqs - nseq) bytes, which always dominates)
+    u8* lit_buf = reinterpret_cast<u8*>(seqs + nseq);
+    {
       u64 p = block_off;
-      u64 w = lp;
+      u64 lw = 0;
       for (u32 i = 0; i < nseq; i++) {
-        mx_par_copy(out + w, src + p, seqs[i].ll);
-        w += seqs[i].ll;
-        p += seqs[i].ll + seqs[i].ml;
+        mx_par_copy(lit_buf + lw, src + p, seqs[i].ll);
+        lw += seqs[i].ll;
+        p += seqs[i].ll + (u64)seqs[i].ml;
       }
-      u64 trail = (block_off + block_len) - p;
-      mx_par_copy(out + w, src + p, trail);
-      w += trail;
-      // nseq header
+      mx_par_copy(lit_buf + lw, src + p, (block_off + block_len) - p);
+    }
+    lit_histogram(lit_buf, lit_total, et->lit_hist);
+    u32 nsym = 0;
+    for (u32 i = 0; i < 256; i++)
+      if (et->lit_hist[i]) nsym++;
+
+    // ---- literals section: RLE / huffman / raw ----
+    u64 w = 0;
+    bool lit_done = false;
+    if (lit_total > 0 && nsym == 1) {  // RLE literals
+      if (lit_total <= 31) {
+        out[w++] = (u8)(1 | (0 << 2) | (lit_total << 3));
+      } else if (lit_total <= 4095) {
+        out[w++] = (u8)(1 | (1 << 2) | ((lit_total & 0xF) << 4));
+        out[w++] = (u8)(lit_total >> 4);
+      } else {
+        out[w++] = (u8)(1 | (3 << 2) | ((lit_total & 0xF) << 4));
+        out[w++] = (u8)((lit_total >> 4) & 0xFF);
+        out[w++] = (u8)(lit_total >> 12);
+      }
+      out[w++] = lit_buf[0];
+      lit_done = true;
+    }
+    if (!lit_done && lit_total >= 512 && nsym > 1 &&
+        huf_build_enc(et->lit_hist, lit_total, &et->he)) {
+      i64 n = emit_huf_literals(&et->he, lit_buf, lit_total, out + w, cap - w - 32);
+      if (n > 0) {
+        w += (u64)n;
+        lit_done = true;
+      }
+    }
+    if (!lit_done) {  // raw literals
+      u64 lp = w;
+      if (lit_total <= 31) {
+        out[lp++] = (u8)(0 | (0 << 2) | (lit_total << 3));
+      } else if (lit_total <= 4095) {
+        out[lp++] = (u8)(0 | (1 << 2) | ((lit_total & 0xF) << 4));
+        out[lp++] = (u8)(lit_total >> 4);
+      } else {
+        out[lp++] = (u8)(0 | (3 << 2) | ((lit_total & 0xF) << 4));
+        out[lp++] = (u8)((lit_total >> 4) & 0xFF);
+        out[lp++] = (u8)(lit_total >> 12);
+      }
+      if (lp + lit_total + 16 > cap) {
+        use_raw = true;
+      } else {
+        mx_par_copy(out + lp, lit_buf, lit_total);
+        w = lp + lit_total;
+      }
+    }
+
+    if (!use_raw) {
+      // ---- sequences section ----
       if (nseq < 128) {
         out[w++] = (u8)nseq;
       } else if (nseq < 0x7F00) {
@@ -349,9 +621,11 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
         out[w++] = (u8)((nseq - 0x7F00) >> 8);
       }
       out[w++] = 0;  // modes: all predefined
-      // sequence bitstream (encode in reverse)
+      const FseEnc* ell = &et->ell;
+      const FseEnc* eof = &et->eof;
+      const FseEnc* eml = &et->eml;
       BitW bw;
-      bw.init(out + w, cap - w > block_len ? block_len : cap - w);  // bounded: must beat raw
+      bw.init(out + w, cap - w > block_len ? block_len : cap - w);  // must beat raw
       const Seq& lastq = seqs[nseq - 1];
       u32 ll_c = ll_code_of(lastq.ll);
       u32 ml_c = ml_code_of(lastq.ml);
@@ -411,30 +685,13 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
   return (i64)(3 + csize);
 }
 
-// Predefined-table FSE encoders (caller-allocated: LDS on the GPU so the
-// redundant-wavefront execution doesn't spill per-lane copies).
-struct EncTables {
-  FseEnc ell, eof, eml;
-};
-
-MX_HD static inline int enc_tables_init(EncTables* et) {
-  u32 nsym, log;
-  const i16* d = ll_default_dist(&nsym, &log);
-  if (fse_build_ctable(&et->ell, d, nsym, log) < 0) return MXZ_ERR_FSE;
-  d = of_default_dist(&nsym, &log);
-  if (fse_build_ctable(&et->eof, d, nsym, log) < 0) return MXZ_ERR_FSE;
-  d = ml_default_dist(&nsym, &log);
-  if (fse_build_ctable(&et->eml, d, nsym, log) < 0) return MXZ_ERR_FSE;
-  return MXZ_OK;
-}
 
 // Encode src[0..len) as ONE standard zstd frame into dst. `hash` is a
 // (1<<kHashLog) u32 scratch (will be zeroed), `seqs` holds >= kBlockMax/4+1
 // entries, `et` holds initialized predefined tables (enc_tables_init).
 // Returns frame size or <0.
 MX_HD static inline i64 encode_frame(const u8* src, u64 len, u8* dst, u64 dstcap, u32* hash,
-                                     Seq* seqs, const EncTables* et) {
-  const FseEnc &ell = et->ell, &eof = et->eof, &eml = et->eml;
+                                     Seq* seqs, EncTables* et) {
 
   {
     u32 lane = mx_lane(), w = mx_width();
@@ -488,7 +745,7 @@ MX_HD static inline i64 encode_frame(const u8* src, u64 len, u8* dst, u64 dstcap
     u64 blen = len - off < kBlockMax ? len - off : kBlockMax;
     bool last = off + blen >= len;
     i64 n = encode_block(src, off, blen, last, dst + pos, dstcap - pos, hash, seqs,
-                         (u32)(kBlockMax / 4 + 1), &ell, &eof, &eml);
+                         (u32)(kBlockMax / 4 + 1), et);
     if (n < 0) return n;
     pos += (u64)n;
     off += blen;

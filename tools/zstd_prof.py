@@ -22,6 +22,8 @@ from modelx_amd import _core
 def payload(kind: str, nbytes: int) -> torch.Tensor:
     if kind == "random":
         return torch.randint(0, 256, (nbytes,), dtype=torch.uint8, device="cuda")
+    if kind == "low-entropy":  # 64 symbols uniform: pure huffman-literals work
+        return torch.randint(0, 64, (nbytes,), dtype=torch.uint8, device="cuda")
     if kind == "tiled4k":
         page = torch.randint(0, 256, (4096,), dtype=torch.uint8, device="cuda")
         return page.repeat(nbytes // 4096 + 1)[:nbytes].contiguous()
@@ -41,7 +43,7 @@ def main():
     eng = _core.GpuEngine(device=0, num_slots=4, slot_bytes=8 << 20, num_streams=2)
     frame_raw = args.frame_kib << 10
 
-    for kind in ("tiled4k", "text", "random"):
+    for kind in ("tiled4k", "text", "low-entropy", "random"):
         src = payload(kind, n)
         bound = _core.zstd_compress_bound(n, frame_raw)
         comp = torch.empty(bound, dtype=torch.uint8, device="cuda")
