@@ -549,12 +549,14 @@ struct DecCtx {
   FseTable ll, of, ml;  // persist across blocks (repeat mode)
   BuildScratch bs;      // table-construction temporaries
   bool ll_valid, of_valid, ml_valid;
+  int stream_rc;        // per-wave huffman stream results (multi-wave decode)
   u8* lit_scratch;      // >= kBlockMax bytes
 };
 
 MX_HD static inline void dec_ctx_init(DecCtx* c, u8* lit_scratch) {
   c->huf.valid = false;
   c->ll_valid = c->of_valid = c->ml_valid = false;
+  c->stream_rc = MXZ_OK;
   c->lit_scratch = lit_scratch;
 }
 
@@ -696,12 +698,33 @@ MX_HD static inline i64 decode_block(DecCtx* c, u32* rep, const u8* src, u64 src
       if (r123 * 3 > lit_regen) return MXZ_ERR_LITERALS;
       u64 r4 = lit_regen - 3 * r123;
       const u8* sp = cl + 6;
-      int rc = huf_decode_stream(&c->huf, sp, s1, c->lit_scratch, r123);
-      if (rc == MXZ_OK) rc = huf_decode_stream(&c->huf, sp + s1, s2, c->lit_scratch + r123, r123);
-      if (rc == MXZ_OK)
-        rc = huf_decode_stream(&c->huf, sp + s1 + s2, s3, c->lit_scratch + 2 * r123, r123);
-      if (rc == MXZ_OK)
-        rc = huf_decode_stream(&c->huf, sp + s1 + s2 + s3, s4, c->lit_scratch + 3 * r123, r4);
+      const u8* srcs[4] = {sp, sp + s1, sp + s1 + s2, sp + s1 + s2 + s3};
+      u64 slens[4] = {s1, s2, s3, s4};
+      u64 rlens[4] = {r123, r123, r123, r4};
+      int rc = MXZ_OK;
+#if defined(__HIP_DEVICE_COMPILE__)
+      // one wave per stream: the four backward streams are independent,
+      // and huffman decode is the serial tail of literal-heavy frames.
+      // Divergent across waves (no barriers inside), re-converged below.
+      u32 wave = mx_lane() / 64;
+      u32 nwaves = (mx_width() + 63) / 64;
+      int my_rc = MXZ_OK;
+      for (u32 k = wave; k < 4; k += nwaves)
+        if (my_rc == MXZ_OK)
+          my_rc = huf_decode_stream(&c->huf, srcs[k], slens[k],
+                                    c->lit_scratch + (u64)k * r123, rlens[k]);
+      // publish failures (same-value or benign-differing error codes)
+      if (my_rc != MXZ_OK) c->stream_rc = my_rc;
+      mx_sync();
+      rc = c->stream_rc;
+      mx_sync();
+      c->stream_rc = MXZ_OK;  // reset for the next block (all waves write)
+      mx_sync();
+#else
+      for (u32 k = 0; k < 4 && rc == MXZ_OK; k++)
+        rc = huf_decode_stream(&c->huf, srcs[k], slens[k],
+                               c->lit_scratch + (u64)k * r123, rlens[k]);
+#endif
       if (rc < 0) return rc;
     }
     lits = c->lit_scratch;

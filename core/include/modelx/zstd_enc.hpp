@@ -397,8 +397,63 @@ MX_HD static inline bool huf_build_enc(const u32* hist, u64 total, HufEnc* e) {
 }
 
 // Encode lit[a..b) as one backward huffman stream; returns bytes or <0.
+//
+// Device path: the bit position of every code is a prefix sum of code
+// lengths, so the whole stream is written lane-parallel — scan 64 lengths
+// per step (wave shuffles), then each lane ORs its code into the output
+// words (atomicOr; codes are <= 11 bits so they span at most two u32s).
+// The serial BitW path remains for the CPU and for short runs.
 MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a, u64 b,
                                           u8* dst, u64 cap) {
+#if defined(__HIP_DEVICE_COMPILE__)
+  u64 n = b - a;
+  if (n >= 256) {
+    u32 lane = threadIdx.x & 63;
+    // total bits (parallel reduction)
+    u64 my_bits = 0;
+    for (u64 k = lane; k < n; k += 64) my_bits += e->len[lit[b - 1 - k]];
+    for (int d = 32; d >= 1; d >>= 1) my_bits += __shfl_down((unsigned long long)my_bits, d);
+    u64 T = __shfl((unsigned long long)my_bits, 0);
+    u64 bytes = (T + 1 + 7) / 8;
+    if (bytes > cap) return MXZ_ERR_DST_SMALL;
+    // word-aligned view; preserve the bytes before dst in the first word
+    uintptr_t addr = reinterpret_cast<uintptr_t>(dst);
+    u32 misal = (u32)(addr & 3);
+    u32* words = reinterpret_cast<u32*>(addr - misal);
+    u64 bit0 = (u64)misal * 8;  // stream bit i lives at word bit (bit0 + i)
+    u64 nwords = (bit0 + T + 1 + 31) / 32;
+    u32 keep = 0;
+    if (lane == 0 && misal) {
+      for (u32 i = 0; i < misal; i++) keep |= (u32)dst[-(i32)misal + i] << (8 * i);
+    }
+    for (u64 wdi = lane; wdi < nwords; wdi += 64) words[wdi] = 0;
+    mx_sync();
+    if (lane == 0 && misal) atomicOr(&words[0], keep);
+    // chunked scan + scatter
+    u64 running = 0;
+    for (u64 base = 0; base < n; base += 64) {
+      u64 k = base + lane;
+      u32 sym = k < n ? lit[b - 1 - k] : 0;
+      u32 nk = k < n ? e->len[sym] : 0;
+      u32 inc = nk;
+      for (int d = 1; d < 64; d <<= 1) {
+        u32 up = __shfl_up(inc, d);
+        if ((int)lane >= d) inc += up;
+      }
+      u64 S = running + inc - nk + bit0;
+      if (k < n && nk) {
+        u64 wide = (u64)e->code[sym] << (S & 31);
+        atomicOr(&words[S >> 5], (u32)wide);
+        u32 hi = (u32)(wide >> 32);
+        if (hi) atomicOr(&words[(S >> 5) + 1], hi);
+      }
+      running += __shfl(inc, 63);
+    }
+    if (lane == 0) atomicOr(&words[(bit0 + T) >> 5], 1u << ((bit0 + T) & 31));
+    mx_sync();
+    return (i64)bytes;
+  }
+#endif
   BitW bw;
   bw.init(dst, cap);
   // written back-to-front so the backward reader produces them in order
