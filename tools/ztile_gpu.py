@@ -18,11 +18,18 @@ def main():
     z.ZSTD_decompress.restype = ctypes.c_size_t
     z.ZSTD_isError.restype = ctypes.c_uint
 
-    page = torch.randint(0, 256, (64 << 10,), dtype=torch.uint8, device="cuda:0")
+    import sys as _sys
+
+    kind = _sys.argv[1] if len(_sys.argv) > 1 else "tiled"
     n = 64 << 20
-    src = page.repeat(n // page.numel())[:n].contiguous()
-    torch.manual_seed(1000)
-    src[: 4 << 20] = torch.randint(0, 256, (4 << 20,), dtype=torch.uint8, device="cuda:0")
+    if kind == "lowent":
+        src = torch.randint(0, 64, (n,), dtype=torch.uint8, device="cuda:0")
+    else:
+        page = torch.randint(0, 256, (64 << 10,), dtype=torch.uint8, device="cuda:0")
+        src = page.repeat(n // page.numel())[:n].contiguous()
+        torch.manual_seed(1000)
+        src[: 4 << 20] = torch.randint(0, 256, (4 << 20,), dtype=torch.uint8,
+                                       device="cuda:0")
     torch.cuda.synchronize()
 
     bound = _core.zstd_compress_bound(n)
@@ -60,6 +67,16 @@ def main():
                 print(f"frame {i}: lib_ok={lib_ok} cpu_ok={cpu_ok} "
                       f"c_size={c_size} d_size={d_size} -> dumped {path}")
     print(f"bad frames: {bad} / {len(frames)}")
+    # CPU-compress the same content and GPU-decode it (isolates decoder)
+    cpu_blob = _core.zstd_compress_cpu(src_host, 128 << 10)
+    csrc = torch.frombuffer(bytearray(cpu_blob), dtype=torch.uint8).cuda()
+    cback = torch.empty(n, dtype=torch.uint8, device="cuda:0")
+    try:
+        m2 = eng.zstd_decompress_device(csrc.data_ptr(), len(cpu_blob), cback.data_ptr(), n)
+        ok2 = m2 == n and bytes(cback.cpu().numpy().tobytes()) == src_host
+        print("gpu decode of CPU blob:", "OK" if ok2 else "MISMATCH")
+    except Exception as e:
+        print("gpu decode of CPU blob FAILED:", e)
     # also exercise GPU decode end-to-end
     back = torch.empty(n, dtype=torch.uint8, device="cuda:0")
     try:
