@@ -32,10 +32,11 @@ struct MxzFrameC {
 // win on raw-literal frames (our encoder's output), which are copy-bound.
 __global__ __launch_bounds__(256) void zstd_decompress_frames_kernel(
     const u8* __restrict__ src, const MxzFrameC* __restrict__ frames, u32 nframes,
-    u8* __restrict__ dst, u8* __restrict__ lit_scratch, i64* __restrict__ rc) {
+    u8* __restrict__ dst, u8* __restrict__ lit_scratch, i64* __restrict__ rc, u32 flags) {
   __shared__ DecCtx ctx;
   u32 f = blockIdx.x;
   if (f >= nframes) return;
+  ctx.flags = flags;
   ctx.lit_scratch = lit_scratch + (u64)f * kBlockMax;
   MxzFrameC fr = frames[f];
   i64 n = decode_frame(src + fr.c_off, fr.c_size, dst + fr.d_off, fr.d_size, &ctx, nullptr);
@@ -45,11 +46,12 @@ __global__ __launch_bounds__(256) void zstd_decompress_frames_kernel(
 extern "C" hipError_t modelx_zstd_decompress_frames(const void* src, const void* frames_dev,
                                                     uint32_t nframes, void* dst,
                                                     void* lit_scratch, int64_t* rc_dev,
-                                                    hipStream_t stream) {
+                                                    uint32_t flags, hipStream_t stream) {
   if (nframes == 0) return hipSuccess;
   hipLaunchKernelGGL(zstd_decompress_frames_kernel, dim3(nframes), dim3(256), 0, stream,
                      static_cast<const u8*>(src), static_cast<const MxzFrameC*>(frames_dev),
-                     nframes, static_cast<u8*>(dst), static_cast<u8*>(lit_scratch), rc_dev);
+                     nframes, static_cast<u8*>(dst), static_cast<u8*>(lit_scratch), rc_dev,
+                     flags);
   return hipGetLastError();
 }
 
@@ -58,9 +60,10 @@ extern "C" hipError_t modelx_zstd_decompress_frames(const void* src, const void*
 __global__ __launch_bounds__(64) void zstd_compress_frames_kernel(
     const u8* __restrict__ src, u64 srclen, u32 frame_raw, u32 first_frame, u32 nframes,
     u8* __restrict__ dst_scratch, u64 stride, Seq* __restrict__ seq_scratch, u32 max_seqs,
-    i64* __restrict__ out_sizes) {
+    i64* __restrict__ out_sizes, u32 flags) {
   __shared__ u32 hash[1u << kHashLog];  // 32 KiB
   __shared__ EncTables et;
+  et.flags = flags;
   u32 b = blockIdx.x;
   if (b >= nframes) return;
   u32 f = first_frame + b;
@@ -82,11 +85,11 @@ extern "C" hipError_t modelx_zstd_compress_frames(const void* src, uint64_t srcl
                                                   uint32_t nframes, void* dst_scratch,
                                                   uint64_t stride, void* seq_scratch,
                                                   uint32_t max_seqs, int64_t* out_sizes_dev,
-                                                  hipStream_t stream) {
+                                                  uint32_t flags, hipStream_t stream) {
   if (nframes == 0) return hipSuccess;
   hipLaunchKernelGGL(zstd_compress_frames_kernel, dim3(nframes), dim3(64), 0, stream,
                      static_cast<const u8*>(src), srclen, frame_raw, first_frame, nframes,
                      static_cast<u8*>(dst_scratch), stride, static_cast<Seq*>(seq_scratch),
-                     max_seqs, out_sizes_dev);
+                     max_seqs, out_sizes_dev, flags);
   return hipGetLastError();
 }

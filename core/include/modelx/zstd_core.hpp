@@ -545,6 +545,7 @@ MX_HD static inline const i16* of_default_dist(u32* nsym, u32* log) {
 // -------------------------------------------------------- decode context ---
 
 struct DecCtx {
+  u32 flags;            // bit1: disable wave-split literal-stream decode
   HufTable huf;         // persists across blocks (treeless literals)
   FseTable ll, of, ml;  // persist across blocks (repeat mode)
   BuildScratch bs;      // table-construction temporaries
@@ -703,23 +704,29 @@ MX_HD static inline i64 decode_block(DecCtx* c, u32* rep, const u8* src, u64 src
       u64 rlens[4] = {r123, r123, r123, r4};
       int rc = MXZ_OK;
 #if defined(__HIP_DEVICE_COMPILE__)
-      // one wave per stream: the four backward streams are independent,
-      // and huffman decode is the serial tail of literal-heavy frames.
-      // Divergent across waves (no barriers inside), re-converged below.
-      u32 wave = mx_lane() / 64;
-      u32 nwaves = (mx_width() + 63) / 64;
-      int my_rc = MXZ_OK;
-      for (u32 k = wave; k < 4; k += nwaves)
-        if (my_rc == MXZ_OK)
-          my_rc = huf_decode_stream(&c->huf, srcs[k], slens[k],
-                                    c->lit_scratch + (u64)k * r123, rlens[k]);
-      // publish failures (same-value or benign-differing error codes)
-      if (my_rc != MXZ_OK) c->stream_rc = my_rc;
-      mx_sync();
-      rc = c->stream_rc;
-      mx_sync();
-      c->stream_rc = MXZ_OK;  // reset for the next block (all waves write)
-      mx_sync();
+      if (c->flags & 2u) {
+        for (u32 k = 0; k < 4 && rc == MXZ_OK; k++)
+          rc = huf_decode_stream(&c->huf, srcs[k], slens[k],
+                                 c->lit_scratch + (u64)k * r123, rlens[k]);
+      } else {
+        // one wave per stream: the four backward streams are independent,
+        // and huffman decode is the serial tail of literal-heavy frames.
+        // Divergent across waves (no barriers inside), re-converged below.
+        u32 wave = mx_lane() / 64;
+        u32 nwaves = (mx_width() + 63) / 64;
+        int my_rc = MXZ_OK;
+        for (u32 k = wave; k < 4; k += nwaves)
+          if (my_rc == MXZ_OK)
+            my_rc = huf_decode_stream(&c->huf, srcs[k], slens[k],
+                                      c->lit_scratch + (u64)k * r123, rlens[k]);
+        // publish failures (same-value or benign-differing error codes)
+        if (my_rc != MXZ_OK) c->stream_rc = my_rc;
+        mx_sync();
+        rc = c->stream_rc;
+        mx_sync();
+        c->stream_rc = MXZ_OK;  // reset for the next block (all waves write)
+        mx_sync();
+      }
 #else
       for (u32 k = 0; k < 4 && rc == MXZ_OK; k++)
         rc = huf_decode_stream(&c->huf, srcs[k], slens[k],

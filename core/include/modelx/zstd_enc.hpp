@@ -404,10 +404,10 @@ MX_HD static inline bool huf_build_enc(const u32* hist, u64 total, HufEnc* e) {
 // words (atomicOr; codes are <= 11 bits so they span at most two u32s).
 // The serial BitW path remains for the CPU and for short runs.
 MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a, u64 b,
-                                          u8* dst, u64 cap) {
+                                          u8* dst, u64 cap, u32 flags) {
 #if defined(__HIP_DEVICE_COMPILE__)
   u64 n = b - a;
-  if (n >= 256) {
+  if (n >= 256 && !(flags & 1u)) {
     u32 lane = threadIdx.x & 63;
     // total bits (parallel reduction)
     u64 my_bits = 0;
@@ -471,7 +471,7 @@ MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a,
 // table). Returns total section bytes (header included) or <0 when raw is
 // better / capacity exceeded.
 MX_HD static inline i64 emit_huf_literals(const HufEnc* e, const u8* lit, u64 n, u8* out,
-                                          u64 cap) {
+                                          u64 cap, u32 flags) {
   if (n < 256) return MXZ_ERR_DST_SMALL;  // not worth the table
   // table: header byte + packed 4-bit weights for symbols 0..last_sym-1
   u32 nweights = e->last_sym;  // last symbol's weight is implied
@@ -503,7 +503,8 @@ MX_HD static inline i64 emit_huf_literals(const HufEnc* e, const u8* lit, u64 n,
   u64 offs[4] = {0, r123, 2 * r123, 3 * r123};
   u64 lens[4] = {r123, r123, r123, r4};
   for (int k = 0; k < 4; k++) {
-    i64 m = huf_encode_stream(e, lit, offs[k], offs[k] + lens[k], body + bpos, bcap - bpos);
+    i64 m = huf_encode_stream(e, lit, offs[k], offs[k] + lens[k], body + bpos, bcap - bpos,
+                              flags);
     if (m < 0) return m;
     if (k < 3 && m > 0xFFFF) return MXZ_ERR_DST_SMALL;
     sizes[k] = (u64)m;
@@ -534,6 +535,7 @@ MX_HD static inline i64 emit_huf_literals(const HufEnc* e, const u8* lit, u64 n,
 // LDS on the GPU so the redundant-wavefront execution doesn't spill
 // per-lane copies; the histogram is filled with LDS atomics).
 struct EncTables {
+  u32 flags;  // bit0: disable the lane-parallel huffman stream encoder
   FseEnc ell, eof, eml;
   HufEnc he;
   u32 lit_hist[256];
@@ -582,7 +584,7 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
       return 4;
     }
     if (block_len >= 512 && huf_build_enc(et->lit_hist, block_len, &et->he)) {
-      i64 n = emit_huf_literals(&et->he, block, block_len, dst + 3, dstcap - 5);
+      i64 n = emit_huf_literals(&et->he, block, block_len, dst + 3, dstcap - 5, et->flags);
       if (n > 0 && (u64)n + 1 < block_len) {
         dst[3 + n] = 0;  // zero sequences
         csize = (u64)n + 1;
@@ -637,7 +639,8 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
     }
     if (!lit_done && lit_total >= 512 && nsym > 1 &&
         huf_build_enc(et->lit_hist, lit_total, &et->he)) {
-      i64 n = emit_huf_literals(&et->he, lit_buf, lit_total, out + w, cap - w - 32);
+      i64 n = emit_huf_literals(&et->he, lit_buf, lit_total, out + w, cap - w - 32,
+                                et->flags);
       if (n > 0) {
         w += (u64)n;
         lit_done = true;

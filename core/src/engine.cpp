@@ -58,7 +58,7 @@ extern "C" hipError_t modelx_tar_scatter(const void* tar, const void* segs, uint
 extern "C" hipError_t modelx_zstd_decompress_frames(const void* src, const void* frames_dev,
                                                     uint32_t nframes, void* dst,
                                                     void* lit_scratch, int64_t* rc_dev,
-                                                    hipStream_t stream);
+                                                    uint32_t flags, hipStream_t stream);
 extern "C" hipError_t modelx_dedup_insert(const void* leaves_dev, uint32_t nchunks,
                                            uint64_t base, uint64_t chunk_size, uint64_t total,
                                            void* table, uint64_t cap, uint32_t* dropped_dev,
@@ -76,11 +76,22 @@ extern "C" hipError_t modelx_zstd_compress_frames(const void* src, uint64_t srcl
                                                   uint32_t nframes, void* dst_scratch,
                                                   uint64_t stride, void* seq_scratch,
                                                   uint32_t max_seqs, int64_t* out_sizes_dev,
-                                                  hipStream_t stream);
+                                                  uint32_t flags, hipStream_t stream);
 
 struct MxzFrameHost {
   uint64_t c_off, c_size, d_off, d_size;
 };
+
+// MODELX_ZSTD_FLAGS: bit0 disables the lane-parallel huffman encoder,
+// bit1 disables the wave-split literal-stream decoder (bisection/panic
+// switches for the device codec).
+static uint32_t zstd_flags() {
+  static uint32_t f = [] {
+    const char* v = getenv("MODELX_ZSTD_FLAGS");
+    return v ? (uint32_t)atoi(v) : 0u;
+  }();
+  return f;
+}
 
 struct TarEntryHost {
   uint64_t header_off, payload_off, size;
@@ -725,7 +736,7 @@ class GpuEngine {
       uint32_t n = (uint32_t)std::min<uint64_t>(batch, nframes - first);
       HIP_CHECK(modelx_zstd_compress_frames(reinterpret_cast<void*>(src_ptr), size, frame_raw,
                                             (uint32_t)first, n, dscratch, stride, dseqs,
-                                            max_seqs, dsizes, hash_stream_));
+                                            max_seqs, dsizes, zstd_flags(), hash_stream_));
       HIP_CHECK(hipStreamSynchronize(hash_stream_));
       HIP_CHECK(hipMemcpy(hsizes.data(), dsizes, n * sizeof(int64_t), hipMemcpyDeviceToHost));
       for (uint32_t i = 0; i < n; i++) {
@@ -808,7 +819,7 @@ class GpuEngine {
                                hipMemcpyHostToDevice, hash_stream_));
       HIP_CHECK(modelx_zstd_decompress_frames(reinterpret_cast<void*>(src_ptr), dframes, n,
                                               reinterpret_cast<void*>(dst_ptr), dlit, drc,
-                                              hash_stream_));
+                                              zstd_flags(), hash_stream_));
       HIP_CHECK(hipStreamSynchronize(hash_stream_));
       HIP_CHECK(hipMemcpy(hrc.data(), drc, n * sizeof(int64_t), hipMemcpyDeviceToHost));
       for (uint32_t i = 0; i < n; i++)

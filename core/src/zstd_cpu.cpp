@@ -176,6 +176,7 @@ std::vector<uint8_t> compress_seekable(const uint8_t* src, size_t len, uint32_t 
       std::vector<u32> hash(1u << kHashLog);
       std::vector<Seq> seqs(kBlockMax / 4 + 1);
       EncTables et;
+      et.flags = 0;
       enc_tables_init(&et);
       size_t i;
       while ((i = next.fetch_add(1)) < nframes) {
@@ -232,6 +233,7 @@ std::vector<uint8_t> decompress(const uint8_t* blob, size_t len) {
     pool.emplace_back([&] {
       std::vector<u8> lit(kBlockMax);
       auto ctx = std::make_unique<DecCtx>();
+      ctx->flags = 0;
       ctx->lit_scratch = lit.data();
       size_t i;
       while ((i = next.fetch_add(1)) < table.size()) {
