@@ -404,16 +404,20 @@ MX_HD static inline bool huf_build_enc(const u32* hist, u64 total, HufEnc* e) {
 // words (atomicOr; codes are <= 11 bits so they span at most two u32s).
 // The serial BitW path remains for the CPU and for short runs.
 MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a, u64 b,
-                                          u8* dst, u64 cap, u32 flags) {
+                                          u8* dst, u64 cap, u32 flags, u32* scan_tmp) {
 #if defined(__HIP_DEVICE_COMPILE__)
   u64 n = b - a;
   if (n >= 256 && !(flags & 1u)) {
     u32 lane = threadIdx.x & 63;
-    // total bits (parallel reduction)
-    u64 my_bits = 0;
+    // total bits: per-lane partials exchanged through LDS (explicit
+    // barriers — no reliance on wave-shuffle out-of-range semantics)
+    u32 my_bits = 0;
     for (u64 k = lane; k < n; k += 64) my_bits += e->len[lit[b - 1 - k]];
-    for (int d = 32; d >= 1; d >>= 1) my_bits += __shfl_down((unsigned long long)my_bits, d);
-    u64 T = __shfl((unsigned long long)my_bits, 0);
+    scan_tmp[lane] = my_bits;
+    mx_sync();
+    u64 T = 0;
+    for (u32 i = 0; i < 64; i++) T += scan_tmp[i];
+    mx_sync();
     u64 bytes = (T + 1 + 7) / 8;
     if (bytes > cap) return MXZ_ERR_DST_SMALL;
     // word-aligned view; preserve the bytes before dst in the first word
@@ -429,25 +433,30 @@ MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a,
     for (u64 wdi = lane; wdi < nwords; wdi += 64) words[wdi] = 0;
     mx_sync();
     if (lane == 0 && misal) atomicOr(&words[0], keep);
-    // chunked scan + scatter
+    // chunked scan + scatter: lengths exchanged through LDS, every lane
+    // computes its own exclusive prefix
     u64 running = 0;
     for (u64 base = 0; base < n; base += 64) {
       u64 k = base + lane;
       u32 sym = k < n ? lit[b - 1 - k] : 0;
       u32 nk = k < n ? e->len[sym] : 0;
-      u32 inc = nk;
-      for (int d = 1; d < 64; d <<= 1) {
-        u32 up = __shfl_up(inc, d);
-        if ((int)lane >= d) inc += up;
+      scan_tmp[lane] = nk;
+      mx_sync();
+      u32 excl = 0;
+      u32 chunk_total = 0;
+      for (u32 i = 0; i < 64; i++) {
+        if (i < lane) excl += scan_tmp[i];
+        chunk_total += scan_tmp[i];
       }
-      u64 S = running + inc - nk + bit0;
+      mx_sync();
+      u64 S = running + excl + bit0;
       if (k < n && nk) {
         u64 wide = (u64)e->code[sym] << (S & 31);
         atomicOr(&words[S >> 5], (u32)wide);
         u32 hi = (u32)(wide >> 32);
         if (hi) atomicOr(&words[(S >> 5) + 1], hi);
       }
-      running += __shfl(inc, 63);
+      running += chunk_total;
     }
     if (lane == 0) atomicOr(&words[(bit0 + T) >> 5], 1u << ((bit0 + T) & 31));
     mx_sync();
@@ -471,7 +480,7 @@ MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a,
 // table). Returns total section bytes (header included) or <0 when raw is
 // better / capacity exceeded.
 MX_HD static inline i64 emit_huf_literals(const HufEnc* e, const u8* lit, u64 n, u8* out,
-                                          u64 cap, u32 flags) {
+                                          u64 cap, u32 flags, u32* scan_tmp) {
   if (n < 256) return MXZ_ERR_DST_SMALL;  // not worth the table
   // table: header byte + packed 4-bit weights for symbols 0..last_sym-1
   u32 nweights = e->last_sym;  // last symbol's weight is implied
@@ -504,7 +513,7 @@ MX_HD static inline i64 emit_huf_literals(const HufEnc* e, const u8* lit, u64 n,
   u64 lens[4] = {r123, r123, r123, r4};
   for (int k = 0; k < 4; k++) {
     i64 m = huf_encode_stream(e, lit, offs[k], offs[k] + lens[k], body + bpos, bcap - bpos,
-                              flags);
+                              flags, scan_tmp);
     if (m < 0) return m;
     if (k < 3 && m > 0xFFFF) return MXZ_ERR_DST_SMALL;
     sizes[k] = (u64)m;
@@ -539,6 +548,7 @@ struct EncTables {
   FseEnc ell, eof, eml;
   HufEnc he;
   u32 lit_hist[256];
+  u32 scan_tmp[64];  // cross-lane exchange for the parallel huffman encoder
 };
 
 MX_HD static inline int enc_tables_init(EncTables* et) {
@@ -584,7 +594,8 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
       return 4;
     }
     if (block_len >= 512 && huf_build_enc(et->lit_hist, block_len, &et->he)) {
-      i64 n = emit_huf_literals(&et->he, block, block_len, dst + 3, dstcap - 5, et->flags);
+      i64 n = emit_huf_literals(&et->he, block, block_len, dst + 3, dstcap - 5, et->flags,
+                                et->scan_tmp);
       if (n > 0 && (u64)n + 1 < block_len) {
         dst[3 + n] = 0;  // zero sequences
         csize = (u64)n + 1;
@@ -640,7 +651,7 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
     if (!lit_done && lit_total >= 512 && nsym > 1 &&
         huf_build_enc(et->lit_hist, lit_total, &et->he)) {
       i64 n = emit_huf_literals(&et->he, lit_buf, lit_total, out + w, cap - w - 32,
-                                et->flags);
+                                et->flags, et->scan_tmp);
       if (n > 0) {
         w += (u64)n;
         lit_done = true;
