@@ -428,7 +428,11 @@ MX_HD static inline i64 huf_encode_stream(const HufEnc* e, const u8* lit, u64 a,
     u64 nwords = (bit0 + T + 1 + 31) / 32;
     u32 keep = 0;
     if (lane == 0 && misal) {
-      for (u32 i = 0; i < misal; i++) keep |= (u32)dst[-(i32)misal + i] << (8 * i);
+      // NB: index arithmetic kept in signed/pointer space — mixing a
+      // negated value with an unsigned index promotes to u32 and wraps to
+      // a ~4 GiB offset (page-faulted on hardware)
+      const u8* before = dst - misal;
+      for (u32 i = 0; i < misal; i++) keep |= (u32)before[i] << (8 * i);
     }
     for (u64 wdi = lane; wdi < nwords; wdi += 64) words[wdi] = 0;
     mx_sync();
