@@ -71,6 +71,8 @@ def _payloads():
         "zeros": bytes(150_000),
         "repeats": bytes(reps),
         "low-entropy": bytes(rng.choice(b"aab") for _ in range(140_000)),
+        # 64-symbol uniform: no LZ matches -> pure huffman-literal blocks
+        "entropy-only": bytes(rng.randrange(64) for _ in range(300_000)),
     }
 
 
@@ -95,6 +97,16 @@ class TestCpuCodec:
         data = b"modelx " * 100_000
         blob = _core.zstd_compress_cpu(data, 128 << 10)
         assert len(blob) < len(data) // 20
+
+    def test_huffman_literals_engage(self):
+        import random
+
+        rng = random.Random(5)
+        data = bytes(rng.randrange(64) for _ in range(400_000))  # 6 bits/byte
+        blob = _core.zstd_compress_cpu(data, 128 << 10)
+        # entropy bound is 0.75; huffman-literal blocks should land near it
+        assert len(blob) < int(len(data) * 0.85), len(blob)
+        assert _core.zstd_decompress_cpu(blob) == data
 
     def test_our_encoder_libzstd_decodes(self, libzstd):
         for name, data in _payloads().items():
