@@ -90,6 +90,50 @@ def cpu_smoke(args):
         shutil.rmtree(work, ignore_errors=True)
 
 
+class _CpuBenchClient:
+    """Transfer-client stand-in for MODELX_BENCH_CPU dry-runs: same call
+    surface as GpuClient for the bench loop, CPU bytes through the normal
+    presigned extension path."""
+
+    def __init__(self, url):
+        from modelx_amd.client import Client
+
+        self._client = Client(url)
+        self.remote = self._client.remote
+        self.last_stats = []
+
+    def push_from_gpu(self, repository, version, tensors, part_bytes=0):
+        import hashlib
+        import tempfile
+
+        d = tempfile.mkdtemp(prefix="bench-cpu-")
+        from modelx_amd.config import ModelConfig
+
+        with open(os.path.join(d, "modelx.yaml"), "w") as f:
+            f.write(ModelConfig(description="bench").to_yaml())
+        for name, t in tensors.items():
+            with open(os.path.join(d, name), "wb") as f:
+                f.write(t.numpy().tobytes())
+        self._client.push(repository, version, d, quiet=True)
+        shutil.rmtree(d, ignore_errors=True)
+
+    def pull_to_gpu(self, repository, version, verify=True):
+        import tempfile
+
+        d = tempfile.mkdtemp(prefix="bench-cpu-out-")
+        self._client.pull(repository, version, d, quiet=True)
+        out = {}
+        import torch
+
+        for name in os.listdir(d):
+            p2 = os.path.join(d, name)
+            if os.path.isfile(p2) and name != "modelx.yaml":
+                with open(p2, "rb") as f:
+                    out[name] = torch.frombuffer(bytearray(f.read()), dtype=torch.uint8)
+        shutil.rmtree(d, ignore_errors=True)
+        return out
+
+
 def main():
     args = parse_args()
     if args.cpu_smoke:
@@ -97,6 +141,13 @@ def main():
         return
 
     import torch
+
+    # MODELX_BENCH_CPU=1: exercise the FULL multi-rank orchestration (rank/
+    # port/store layout, barriers, MAX-reduce, JSON contract) on CPU with
+    # gloo and a fake transfer client — the dry-run for the driver's
+    # torchrun invocation on 8-GPU nodes (tests/test_bench_dist.py). The
+    # normal path is untouched.
+    cpu_mode = os.environ.get("MODELX_BENCH_CPU") == "1"
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
@@ -106,10 +157,12 @@ def main():
     if distributed:
         import torch.distributed as dist
 
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group(backend="nccl")  # RCCL over xGMI
+        if not cpu_mode:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend="gloo" if cpu_mode else "nccl")  # nccl IS RCCL
     device = local_rank
-    torch.cuda.set_device(device)
+    if not cpu_mode:
+        torch.cuda.set_device(device)
 
     # --- per-rank S3 + registry stack -------------------------------------
     # Each rank runs its own modelxd + s3d shard (object stores scale
@@ -144,14 +197,17 @@ def main():
 
         dist.barrier()
 
-    from modelx_amd.client.gpu import GpuClient
-    from modelx_amd.client.registry import RegistryClient
+    if cpu_mode:
+        g = _CpuBenchClient(f"http://127.0.0.1:{mdx_port}")
+    else:
+        from modelx_amd.client.gpu import GpuClient
 
-    g = GpuClient(f"http://127.0.0.1:{mdx_port}", device=device, num_slots=args.slots,
-                  slot_bytes=args.slot_mib << 20, num_conns=args.conns)
+        g = GpuClient(f"http://127.0.0.1:{mdx_port}", device=device, num_slots=args.slots,
+                      slot_bytes=args.slot_mib << 20, num_conns=args.conns)
 
     blob_bytes = int(args.blob_gib * (1 << 30))
-    src = torch.empty(blob_bytes, dtype=torch.uint8, device=f"cuda:{device}")
+    src = torch.empty(blob_bytes, dtype=torch.uint8,
+                      device="cpu" if cpu_mode else f"cuda:{device}")
     repo = f"bench/rank{rank}"
 
     substeps = []
@@ -160,7 +216,8 @@ def main():
         # fresh random payload → no HEAD-dedup shortcut; new digest every step
         t0 = time.monotonic()
         src.random_(0, 256)
-        torch.cuda.synchronize(device)
+        if not cpu_mode:
+            torch.cuda.synchronize(device)
         t1 = time.monotonic()
         g.push_from_gpu(repo, f"s{step_idx}", {"blob.bin": src},
                         part_bytes=args.part_mib << 20)
@@ -173,12 +230,14 @@ def main():
         substeps.append((t1 - t0, t2 - t1, t3 - t2, t4 - t3))
 
     def barrier_sync():
-        torch.cuda.synchronize(device)
+        if not cpu_mode:
+            torch.cuda.synchronize(device)
         if distributed:
             import torch.distributed as dist
 
             dist.barrier()
-        torch.cuda.synchronize(device)
+        if not cpu_mode:
+            torch.cuda.synchronize(device)
 
     for w in range(args.warmup):
         one_step(1000 + w)
@@ -193,7 +252,8 @@ def main():
     if distributed:
         import torch.distributed as dist
 
-        t = torch.tensor([elapsed], dtype=torch.float64, device=f"cuda:{device}")
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device="cpu" if cpu_mode else f"cuda:{device}")
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t.item())
 

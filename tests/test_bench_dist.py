@@ -1,0 +1,30 @@
+"""Dry-run of bench.py's multi-rank orchestration under the EXACT torchrun
+invocation the driver uses at round end (one rank per GPU) — on CPU via
+MODELX_BENCH_CPU=1 (gloo, fake transfer client). Validates rank/port/store
+layout, barrier flow, MAX-reduce and the one-line JSON contract for
+world_size 2 without needing 8 GPUs."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_torchrun_world2_cpu_dry_run(tmp_path):
+    env = dict(os.environ, MODELX_BENCH_CPU="1", MASTER_ADDR="127.0.0.1")
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29671", os.path.join(REPO, "bench.py"),
+           "--gpus", "2", "--steps", "2", "--warmup", "1",
+           "--blob-gib", "0.002", "--store", str(tmp_path / "store")]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=300, cwd=REPO, env=env)
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+    lines = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
+    assert len(lines) == 1, f"exactly one JSON line expected, got {lines}"
+    out = json.loads(lines[0])
+    assert out["n_gpus"] == 2
+    assert out["steps"] == 2
+    assert out["scaling"] == "weak"
+    assert out["value"] > 0
+    assert out["metric"].startswith("push+pull GiB/s")
