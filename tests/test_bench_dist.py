@@ -11,19 +11,23 @@ import sys
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
-def test_torchrun_world2_cpu_dry_run(tmp_path):
+import pytest
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_torchrun_world2_cpu_dry_run(tmp_path, world):
     env = dict(os.environ, MODELX_BENCH_CPU="1", MASTER_ADDR="127.0.0.1")
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29671", os.path.join(REPO, "bench.py"),
-           "--gpus", "2", "--steps", "2", "--warmup", "1",
+           "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+           "--master-port", str(29671 + world * 40), os.path.join(REPO, "bench.py"),
+           "--gpus", str(world), "--steps", "2", "--warmup", "1",
            "--blob-gib", "0.002", "--store", str(tmp_path / "store")]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=300, cwd=REPO, env=env)
     assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
     lines = [ln for ln in r.stdout.splitlines() if ln.startswith("{")]
     assert len(lines) == 1, f"exactly one JSON line expected, got {lines}"
     out = json.loads(lines[0])
-    assert out["n_gpus"] == 2
+    assert out["n_gpus"] == world
     assert out["steps"] == 2
     assert out["scaling"] == "weak"
     assert out["value"] > 0
