@@ -103,7 +103,6 @@ class _CpuBenchClient:
         self.last_stats = []
 
     def push_from_gpu(self, repository, version, tensors, part_bytes=0):
-        import hashlib
         import tempfile
 
         d = tempfile.mkdtemp(prefix="bench-cpu-")
