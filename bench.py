@@ -32,8 +32,9 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
-    p.add_argument("--blob-gib", type=float, default=4.0,
-                   help="per-rank blob size per step (GiB)")
+    p.add_argument("--blob-gib", type=float, default=8.0,
+                   help="per-rank blob size per step (GiB; 8 amortizes per-blob "
+                        "control plane, closest bench shape to BASELINE config 2)")
     p.add_argument("--conns", type=int, default=16, help="ranged-GET connections per rank")
     p.add_argument("--slot-mib", type=int, default=64)
     p.add_argument("--slots", type=int, default=16)
