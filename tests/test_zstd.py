@@ -186,3 +186,22 @@ class TestZstdPushPull:
         c.remote.put_manifest("lib/zstd-bad", "v1", manifest)
         with pytest.raises(Exception):
             c.pull("lib/zstd-bad", "v1", str(tmp_path / "out2"), quiet=True)
+
+
+def test_decoder_mutation_fuzz():
+    """Bounded fuzz: random byte mutations of valid blobs must never
+    crash the decoder — every outcome is either a clean error or a
+    decode whose corruption the digest layer would catch."""
+    rng = random.Random(31337)
+    base = _payloads()
+    for name in ("text", "repeats", "entropy-only"):
+        data = base[name]
+        blob = bytearray(_core.zstd_compress_cpu(data, 64 << 10))
+        for _ in range(80):
+            mut = bytearray(blob)
+            for _ in range(rng.randrange(1, 4)):
+                mut[rng.randrange(len(mut))] ^= rng.randrange(1, 256)
+            try:
+                _core.zstd_decompress_cpu(bytes(mut))
+            except RuntimeError:
+                pass  # clean rejection
