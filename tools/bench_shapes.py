@@ -162,16 +162,22 @@ def config5(args, g, dist, rank, world, device):
     sync()
     t0 = time.monotonic()
     landed = 0
+    wall = {}
     for _ in range(args.steps):
         g.clear_chunk_index()
+        w0 = time.monotonic()
         outs = g.pull_many(repo, [f"s{i}" for i in range(small_n)], parallel=args.parallel)
         landed += sum(v.numel() for vs in outs.values() for v in vs.values())
         del outs
+        w1 = time.monotonic()
         for i in range(huge_n):
             out = g.pull_to_gpu(repo, f"h{i}")
             landed += sum(v.numel() for v in out.values())
             del out
+        w2 = time.monotonic()
         torch.cuda.empty_cache()
+        wall["smalls"] = wall.get("smalls", 0.0) + (w1 - w0)
+        wall["huges"] = wall.get("huges", 0.0) + (w2 - w1)
     sync()
     dt = time.monotonic() - t0
     import collections
@@ -183,6 +189,7 @@ def config5(args, g, dist, rank, world, device):
     stages = {k: {"s": round(v[0], 2), "gib": round(v[1] / (1 << 30), 2)}
               for k, v in sorted(agg.items())}
     return {"metric": "config5 mixed-index logical pull GiB/s (zstd+dedup)",
+            "wall": {k: round(v, 2) for k, v in wall.items()},
             "value": round(landed * world / dt / (1 << 30), 3),
             "per_rank_logical_gib": round(landed / (1 << 30), 2),
             "small_blobs": small_n, "huge_gib": round(huge_sz / (1 << 30), 2),
