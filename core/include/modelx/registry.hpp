@@ -61,6 +61,8 @@ class Registry {
                 const std::string& digest);
   void blob_location(http::Request&, http::ResponseWriter&, const std::string& name,
                      const std::string& digest, const std::string& purpose);
+  void pull_plan(http::Request&, http::ResponseWriter&, const std::string& name,
+                 const std::string& ref);
   void garbage_collect(http::Request&, http::ResponseWriter&, const std::string& name);
 
   std::shared_ptr<store::RegistryStore> store_;

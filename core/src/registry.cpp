@@ -145,6 +145,14 @@ void Registry::handle(http::Request& req, http::ResponseWriter& w) {
     w.write_all(405, "method not allowed");
     return;
   }
+  if (seg.size() == 5 && seg[2] == "manifests" && seg[4] == "pull-plan" && m == "GET") {
+    const std::string& ref = seg[3];
+    if (!valid_reference(ref)) {
+      response_error(w, wire::ErrorInfo{404, "NAME_INVALID", "invalid reference", ref});
+      return;
+    }
+    return pull_plan(req, w, name, ref);
+  }
   if (seg.size() == 4 && seg[2] == "manifests") {
     const std::string& ref = seg[3];
     if (!valid_reference(ref)) {
@@ -340,6 +348,94 @@ void Registry::blob_location(http::Request& req, http::ResponseWriter& w, const 
   o["provider"] = json::Value(loc.provider);
   o["purpose"] = json::Value(loc.purpose);
   o["properties"] = loc.properties;
+  response_ok(w, json::Value(std::move(o)));
+}
+
+static const char kB64[] = "ABCDEFGHIJKLMNOPQRSTUVWXYZabcdefghijklmnopqrstuvwxyz0123456789+/";
+
+static std::string base64_encode(const std::string& in) {
+  std::string out;
+  out.reserve((in.size() + 2) / 3 * 4);
+  size_t i = 0;
+  for (; i + 3 <= in.size(); i += 3) {
+    uint32_t v = (uint8_t)in[i] << 16 | (uint8_t)in[i + 1] << 8 | (uint8_t)in[i + 2];
+    out += kB64[v >> 18];
+    out += kB64[(v >> 12) & 63];
+    out += kB64[(v >> 6) & 63];
+    out += kB64[v & 63];
+  }
+  size_t rem = in.size() - i;
+  if (rem == 1) {
+    uint32_t v = (uint8_t)in[i] << 16;
+    out += kB64[v >> 18];
+    out += kB64[(v >> 12) & 63];
+    out += "==";
+  } else if (rem == 2) {
+    uint32_t v = (uint8_t)in[i] << 16 | (uint8_t)in[i + 1] << 8;
+    out += kB64[v >> 18];
+    out += kB64[(v >> 12) & 63];
+    out += kB64[(v >> 6) & 63];
+    out += '=';
+  }
+  return out;
+}
+
+// GET /{name}/manifests/{ref}/pull-plan — one-round-trip pull metadata
+// (no reference counterpart; clients fall back to per-blob calls on 404).
+// Returns the manifest plus, per blob digest: the presigned download
+// location and — when small enough — the inlined leaves sidecar (clients
+// verify it against the annotation digest, so it is not trusted).
+void Registry::pull_plan(http::Request& req, http::ResponseWriter& w, const std::string& name,
+                         const std::string& ref) {
+  constexpr int64_t kInlineLeavesMax = 256 << 10;  // blobs <= 1 GiB at 128 KiB chunks
+  wire::Manifest manifest;
+  if (!store_->GetManifest(name, ref, &manifest)) {
+    response_error(w,
+                   wire::ErrorInfo{404, "MANIFEST_UNKNOWN", "manifest: " + ref + " not found", ""});
+    return;
+  }
+  json::Object blobs;
+  std::vector<const wire::Descriptor*> descs;
+  descs.push_back(&manifest.config);
+  for (auto& d : manifest.blobs) descs.push_back(&d);
+  for (const wire::Descriptor* d : descs) {
+    if (d->size == 0 || d->digest.empty()) continue;
+    if (blobs.contains(d->digest)) continue;
+    json::Object e;
+    std::map<std::string, std::string> props;
+    props["size"] = std::to_string(d->size);
+    props["name"] = d->name;
+    auto loc = store_->GetBlobLocation(name, d->digest, "download", props);
+    if (loc.supported) {
+      metrics_.presign_download_total.fetch_add(1);
+      json::Object lo;
+      lo["provider"] = json::Value(loc.provider);
+      lo["purpose"] = json::Value(loc.purpose);
+      lo["properties"] = loc.properties;
+      e["location"] = json::Value(std::move(lo));
+    }
+    std::string leaves_digest;
+    for (auto& kv : d->annotations)
+      if (kv.first == "modelx.amd/leaves-blob") leaves_digest = kv.second;
+    if (!leaves_digest.empty()) {
+      store::FileMeta meta;
+      auto reader = store_->GetBlob(name, leaves_digest, &meta);
+      if (reader && meta.size > 0 && meta.size <= kInlineLeavesMax) {
+        std::string lv(static_cast<size_t>(meta.size), '\0');
+        size_t got = 0;
+        while (got < lv.size()) {
+          ssize_t r = reader->read(&lv[got], lv.size() - got);
+          if (r <= 0) break;
+          got += static_cast<size_t>(r);
+        }
+        if (got == lv.size()) e["leaves64"] = json::Value(base64_encode(lv));
+      }
+    }
+    blobs[d->digest] = json::Value(std::move(e));
+  }
+  json::Object o;
+  o["manifest"] = manifest.to_json();
+  o["blobs"] = json::Value(std::move(blobs));
   response_ok(w, json::Value(std::move(o)));
 }
 

@@ -138,6 +138,33 @@ class GpuClient:
 
     # -------------------------------------------------------------- pull --
 
+    @staticmethod
+    def _plan_url(entry) -> Optional[Tuple[str, Dict[str, str]]]:
+        """(url, headers) from a pull-plan blob entry, if it has one."""
+        loc = (entry or {}).get("location") or {}
+        parts = (loc.get("properties") or {}).get("parts") or []
+        if not parts:
+            return None
+        return parts[0]["url"], _signed_headers(parts[0])
+
+    @staticmethod
+    def _plan_leaves(entry, desc: types.Descriptor) -> Optional[bytes]:
+        """Inlined leaves from a pull-plan entry, VERIFIED against the
+        descriptor's leaves-blob digest (the plan is untrusted input)."""
+        b64 = (entry or {}).get("leaves64")
+        if not b64:
+            return None
+        import base64
+
+        try:
+            data = base64.b64decode(b64)
+        except Exception:
+            return None
+        ld = desc.annotations.get(types.ANNOTATION_LEAVES_BLOB, "")
+        if not ld or dg.sha256_digest(data) != ld:
+            return None
+        return data
+
     def _expected_leaves(self, repository: str, desc: types.Descriptor) -> Optional[bytes]:
         """Fetch the leaves sidecar blob (per-chunk SHA-256 array) if the
         descriptor carries one."""
@@ -194,7 +221,7 @@ class GpuClient:
         return sum(ln for _, ln in ranges)
 
     def pull_zstd_blob_to_device(self, repository: str, desc: types.Descriptor,
-                                 verify: bool = True) -> "torch.Tensor":
+                                 verify: bool = True, plan_entry=None) -> "torch.Tensor":
         """Pull a +zstd blob: land the compressed stream in HBM (streaming
         SHA-256 verify over the stored bytes), decode every frame in its own
         workgroup (core/hip/zstd.hip), then GPU-verify the uncompressed
@@ -204,7 +231,8 @@ class GpuClient:
 
         import torch
 
-        comp = self.pull_blob_to_device(repository, desc, verify=verify)
+        comp = self.pull_blob_to_device(repository, desc, verify=verify,
+                                        plan_entry=plan_entry)
         raw_size = int(desc.annotations.get(types.ANNOTATION_RAW_SIZE, 0) or 0)
         if not raw_size:
             raise er.ModelxError(er.ErrCode.UNSUPPORTED,
@@ -253,7 +281,7 @@ class GpuClient:
 
     def pull_blob_to_device(self, repository: str, desc: types.Descriptor,
                             tensor=None, verify: bool = True,
-                            resume: bool = False) -> "torch.Tensor":
+                            resume: bool = False, plan_entry=None) -> "torch.Tensor":
         """Land a blob in HBM. With ``resume=True`` and an existing tensor,
         only chunks whose GPU hash mismatches the expected leaves are fetched
         (chunk-level resume/dedup — the reference resumes at whole-blob
@@ -268,12 +296,17 @@ class GpuClient:
         if tensor is None:
             tensor = torch.empty(desc.size, dtype=torch.uint8, device=f"cuda:{self.device}")
         assert tensor.numel() >= desc.size
-        url, headers = self._download_url(repository, desc)
+        planned = self._plan_url(plan_entry)
+        url, headers = planned if planned else self._download_url(repository, desc)
+
+        def expected_leaves():
+            return (self._plan_leaves(plan_entry, desc)
+                    or self._expected_leaves(repository, desc))
         cs = int(desc.annotations.get(types.ANNOTATION_CHUNK_SIZE, 0) or 0) or (
             dg.algo_chunk_size(desc.digest.split(":", 1)[0]) or DEFAULT_GPU_CHUNK)
 
         if resume and had_tensor:
-            expect = self._expected_leaves(repository, desc)
+            expect = expected_leaves()
             if expect is not None:
                 got = self.engine.sha256_chunk_leaves(tensor.data_ptr(), desc.size, cs)
                 ranges = self._bad_chunk_ranges(got, expect, cs, desc.size)
@@ -290,7 +323,7 @@ class GpuClient:
             # chunks D2D, fetch only the rest (SURVEY.md §2.2
             # chunk_verify_dedup; the reference dedups at whole-blob
             # granularity only, push.go:169-177)
-            expect = self._expected_leaves(repository, desc)
+            expect = expected_leaves()
             if expect is not None:
                 t0 = time.monotonic()
                 missing, dedup_bytes = self.engine.dedup_pull(
@@ -376,7 +409,7 @@ class GpuClient:
             self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
         except er.ModelxError as first_err:
             # chunk-level refetch before giving up
-            expect = self._expected_leaves(repository, desc)
+            expect = expected_leaves()
             if expect is None:
                 raise er.ModelxError(
                     er.ErrCode.DIGEST_INVALID,
@@ -412,15 +445,27 @@ class GpuClient:
         blobs are landed as raw archive bytes under their blob name.
         ``parallel`` > 1 pulls blobs concurrently (reentrant engine; right
         for many-shard manifests where per-blob latency would stack)."""
-        manifest = self.remote.get_manifest(repository, version)
+        # one round trip for manifest + presigns + leaves when the server
+        # supports pull plans (measured: the per-blob control plane
+        # dominated many-small-blob indexes — docs/roadmap.md)
+        plan = self.remote.get_pull_plan(repository, version)
+        if plan and plan.get("manifest"):
+            manifest = types.Manifest.from_dict(plan["manifest"])
+            plan_blobs = plan.get("blobs") or {}
+        else:
+            manifest = self.remote.get_manifest(repository, version)
+            plan_blobs = {}
         descs = [d for d in manifest.blobs
                  if d.size and d.media_type != types.MEDIA_TYPE_MODEL_LEAVES]
 
         def one(desc):
+            entry = plan_blobs.get(desc.digest)
             if desc.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
                 return desc.name, self.pull_zstd_blob_to_device(repository, desc,
-                                                                verify=verify)
-            return desc.name, self.pull_blob_to_device(repository, desc, verify=verify)
+                                                                verify=verify,
+                                                                plan_entry=entry)
+            return desc.name, self.pull_blob_to_device(repository, desc, verify=verify,
+                                                       plan_entry=entry)
 
         if parallel > 1 and len(descs) > 1:
             from concurrent.futures import ThreadPoolExecutor

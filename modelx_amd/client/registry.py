@@ -166,6 +166,22 @@ class RegistryClient:
         self._raise_for(r)
         return types.BlobLocation.from_dict(r.json())
 
+    def get_pull_plan(self, repository: str, version: str = "") -> Optional[Dict[str, Any]]:
+        """One-round-trip pull metadata: manifest + per-blob presigned
+        download locations + (small) inlined leaves sidecars. Returns None
+        when the server doesn't support the endpoint (MI355X-native
+        addition; reference servers 404 here and the client falls back to
+        per-blob calls). Inlined leaves are verified by the caller against
+        the annotation digest — the plan is an optimization, not a trust
+        root."""
+        r = self.session.get(
+            self._url(repository, "manifests", version or "latest", "pull-plan"),
+            headers=self._headers())
+        if r.status_code in (404, 405, 501):
+            return None
+        self._raise_for(r)
+        return r.json()
+
     def garbage_collect(self, repository: str) -> Dict[str, Any]:
         r = self.session.post(self._url(repository, "garbage-collect"), headers=self._headers())
         self._raise_for(r)

@@ -164,3 +164,73 @@ def test_presigned_url_expiry_enforced(tmp_path):
     finally:
         mdx.stop()
         s3d.stop()
+
+
+def test_pull_plan_endpoint(stack, tmp_path):
+    """Pull-plan: one response carrying manifest + presigned locations +
+    verified-inlineable leaves (MI355X addition; measured to remove the
+    per-blob control-plane round trips that dominate small-blob indexes)."""
+    import hashlib
+
+    import requests
+
+    from modelx_amd.wire import digest as dg
+
+    mdx, _ = stack
+    c = Client(mdx.url)
+    payload = os.urandom(300 * 1024)
+    leaves = b"".join(hashlib.sha256(payload[o : o + (128 << 10)]).digest()
+                      for o in range(0, len(payload), 128 << 10))
+    pd = dg.sha256_digest(payload)
+    ld = dg.sha256_digest(leaves)
+    c.remote.upload_blob_content("plan/model", types.Descriptor(
+        name="w.bin", digest=pd, size=len(payload)), payload)
+    c.remote.upload_blob_content("plan/model", types.Descriptor(
+        name="w.bin.leaves", digest=ld, size=len(leaves)), leaves)
+    cfg = b"description: plan\n"
+    cd = dg.sha256_digest(cfg)
+    c.remote.upload_blob_content("plan/model", types.Descriptor(
+        name="modelx.yaml", digest=cd, size=len(cfg)), cfg)
+    from datetime import datetime, timezone
+
+    m = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
+    m.config = types.Descriptor(name="modelx.yaml", digest=cd, size=len(cfg),
+                                media_type=types.MEDIA_TYPE_MODEL_CONFIG_YAML,
+                                modified=datetime.now(timezone.utc))
+    m.blobs = [
+        types.Descriptor(name="w.bin", digest=pd, size=len(payload),
+                         media_type=types.MEDIA_TYPE_MODEL_FILE,
+                         modified=datetime.now(timezone.utc),
+                         annotations={types.ANNOTATION_LEAVES_BLOB: ld,
+                                      types.ANNOTATION_CHUNK_SIZE: str(128 << 10)}),
+        types.Descriptor(name="w.bin.leaves", digest=ld, size=len(leaves),
+                         media_type=types.MEDIA_TYPE_MODEL_LEAVES,
+                         modified=datetime.now(timezone.utc)),
+    ]
+    c.remote.put_manifest("plan/model", "v1", m)
+
+    plan = c.remote.get_pull_plan("plan/model", "v1")
+    assert plan is not None
+    got_m = types.Manifest.from_dict(plan["manifest"])
+    assert [b.name for b in got_m.blobs] == [b.name for b in m.blobs]
+    entry = plan["blobs"][pd]
+    # presigned location is directly fetchable
+    part = entry["location"]["properties"]["parts"][0]
+    r = requests.get(part["url"], headers={"Range": "bytes=0-99"}, timeout=10)
+    assert r.status_code == 206 and r.content == payload[:100]
+    # inlined leaves round-trip and verify
+    import base64
+
+    assert base64.b64decode(entry["leaves64"]) == leaves
+
+    # client-side verification rejects tampered plan leaves
+    from modelx_amd.client.gpu import GpuClient
+
+    desc = m.blobs[0]
+    good = GpuClient._plan_leaves({"leaves64": entry["leaves64"]}, desc)
+    assert good == leaves
+    bad64 = base64.b64encode(b"x" + leaves[1:]).decode()
+    assert GpuClient._plan_leaves({"leaves64": bad64}, desc) is None
+
+    # unknown manifest -> None (client falls back and 404s properly)
+    assert c.remote.get_pull_plan("plan/model", "nope") is None
