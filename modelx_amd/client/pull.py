@@ -35,15 +35,31 @@ class Puller:
     def __init__(self, remote, concurrency: int = PULL_PUSH_CONCURRENCY):
         self.remote = remote
         self.concurrency = concurrency
+        self._plan_blobs: dict = {}
 
     def pull(self, repository: str, version: str, into_dir: str,
              blob_filter=None, quiet: Optional[bool] = None) -> types.Manifest:
         os.makedirs(into_dir, exist_ok=True)
-        manifest = self.remote.get_manifest(repository, version)
+        # one round trip for manifest + presigned locations where the
+        # server supports pull plans (falls back transparently)
+        plan = None
+        try:
+            plan = self.remote.get_pull_plan(repository, version)
+        except Exception:
+            plan = None
+        if plan and plan.get("manifest"):
+            manifest = types.Manifest.from_dict(plan["manifest"])
+            self._plan_blobs = plan.get("blobs") or {}
+        else:
+            manifest = self.remote.get_manifest(repository, version)
+            self._plan_blobs = {}
         descs: List[types.Descriptor] = [manifest.config] + list(manifest.blobs)
         if blob_filter is not None:
             descs = [d for d in descs if blob_filter(d)]
-        self.pull_blobs(repository, descs, into_dir, quiet=quiet)
+        try:
+            self.pull_blobs(repository, descs, into_dir, quiet=quiet)
+        finally:
+            self._plan_blobs = {}
         return manifest
 
     def pull_blobs(self, repository: str, descs: List[types.Descriptor], into_dir: str,
@@ -138,13 +154,22 @@ class Puller:
 
     # -------------------------------------------------------------- blobs --
 
+    def _planned_location(self, desc: types.Descriptor) -> Optional[types.BlobLocation]:
+        entry = self._plan_blobs.get(desc.digest) if self._plan_blobs else None
+        loc = (entry or {}).get("location")
+        if not loc:
+            return None
+        return types.BlobLocation.from_dict(loc)
+
     def pull_blob(self, repository: str, desc: types.Descriptor, dest_path: str,
                   bar: Optional[Bar] = None, verify: bool = True) -> None:
         """Presigned-location download with registry-stream fallback
         (pull.go:206-215), then digest verification (the reference never
         verifies after download — we do, and re-fetch once on mismatch)."""
         for attempt in range(2):
-            location = self.remote.get_blob_location(repository, desc, "download")
+            location = self._planned_location(desc) if attempt == 0 else None
+            if location is None:
+                location = self.remote.get_blob_location(repository, desc, "download")
             if location is not None:
                 extension = ext.get(location.provider)
                 if extension is None:
