@@ -22,6 +22,7 @@
 //
 // The same ring runs push: device → pinned slot (D2H) → multipart PUT parts.
 #include <hip/hip_runtime.h>
+#include <openssl/evp.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
@@ -463,6 +464,89 @@ class GpuEngine {
     return py::bytes(out);
   }
 
+  // Bounded D2H read of device memory (registry-stream push fallback:
+  // slot-sized pieces fed to a streaming HTTP PUT).
+  py::bytes read_device(uintptr_t src_ptr, uint64_t size) {
+    HIP_CHECK(hipSetDevice(device_));
+    std::string out(size, '\0');
+    {
+      py::gil_scoped_release release;
+      HIP_CHECK(hipMemcpy(&out[0], reinterpret_cast<void*>(src_ptr), size,
+                          hipMemcpyDeviceToHost));
+    }
+    return py::bytes(out);
+  }
+
+  // Canonical (wire-compatible) byte-stream SHA-256 of device memory.
+  // Plain SHA-256 is strictly sequential at 64 B granularity, so a GPU lane
+  // is ~4× SLOWER than a SHA-NI CPU core for a single chain — the right
+  // MI355X split is: stream the bytes D2H through the pinned ring
+  // (double-buffered, overlapped) and run the one sequential chain on the
+  // CPU's SHA-NI units (OpenSSL EVP, ~2 GiB/s/core). Multi-blob pushes get
+  // their parallelism across blobs (one call per blob from a thread pool —
+  // the GIL is released for the whole call). Reference semantics:
+  // pkg/client/push.go:149-161 (go-digest streaming sha256).
+  py::bytes sha256_canonical_device(uintptr_t src_ptr, uint64_t size) {
+    HIP_CHECK(hipSetDevice(device_));
+    unsigned char md[32];
+    {
+      py::gil_scoped_release release;
+      EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+      if (!ctx || EVP_DigestInit_ex(ctx, EVP_sha256(), nullptr) != 1) {
+        if (ctx) EVP_MD_CTX_free(ctx);
+        throw std::runtime_error("EVP sha256 init failed");
+      }
+      if (size) {
+        hipStream_t st_a = streams_[push_rr_.fetch_add(1) % streams_.size()];
+        hipStream_t st_b = streams_[push_rr_.fetch_add(1) % streams_.size()];
+        Slot* cur = nullptr;
+        Slot* nxt = nullptr;
+        {
+          std::unique_lock<std::mutex> lk(mu_);
+          cv_free_.wait(lk, [&] { return free_.size() >= 2; });
+          cur = free_.front();
+          free_.pop();
+          nxt = free_.front();
+          free_.pop();
+        }
+        uint64_t off = 0;
+        uint64_t cur_len = std::min<uint64_t>(slot_bytes_, size);
+        hipError_t e = hipMemcpyAsync(cur->host, reinterpret_cast<char*>(src_ptr), cur_len,
+                                      hipMemcpyDeviceToHost, st_a);
+        if (e == hipSuccess) e = hipEventRecord(cur->event, st_a);
+        while (e == hipSuccess && off < size) {
+          uint64_t next_off = off + cur_len;
+          uint64_t next_len =
+              next_off < size ? std::min<uint64_t>(slot_bytes_, size - next_off) : 0;
+          if (next_len) {
+            e = hipMemcpyAsync(nxt->host, reinterpret_cast<char*>(src_ptr) + next_off,
+                               next_len, hipMemcpyDeviceToHost, st_b);
+            if (e == hipSuccess) e = hipEventRecord(nxt->event, st_b);
+            if (e != hipSuccess) break;
+            std::swap(st_a, st_b);
+          }
+          e = hipEventSynchronize(cur->event);
+          if (e != hipSuccess) break;
+          EVP_DigestUpdate(ctx, cur->host, cur_len);
+          std::swap(cur, nxt);
+          off = next_off;
+          cur_len = next_len;
+        }
+        release_slot(cur);
+        release_slot(nxt);
+        if (e != hipSuccess) {
+          EVP_MD_CTX_free(ctx);
+          throw std::runtime_error(std::string("sha256_canonical_device: hip: ") +
+                                   hipGetErrorString(e));
+        }
+      }
+      unsigned int mdlen = 0;
+      EVP_DigestFinal_ex(ctx, md, &mdlen);
+      EVP_MD_CTX_free(ctx);
+    }
+    return py::bytes(reinterpret_cast<char*>(md), 32);
+  }
+
   // -------------------------------------------------------------- push ----
 
   // Upload device memory [src_ptr, src_ptr+size) as the body of `method` to
@@ -794,18 +878,29 @@ class GpuEngine {
     std::vector<uint8_t> traw(tbl);
     HIP_CHECK(hipMemcpy(traw.data(), reinterpret_cast<char*>(src_ptr) + src_len - tbl, tbl,
                         hipMemcpyDeviceToHost));
+    // validate the skippable-frame envelope like the CPU parser
+    // (zstd_cpu.cpp parse_seek_table): a corrupt-but-stored blob must throw
+    // here, never hand out-of-bounds frame offsets to the decode kernel
+    auto rd32 = [](const uint8_t* p) {
+      return (uint32_t)p[0] | ((uint32_t)p[1] << 8) | ((uint32_t)p[2] << 16) |
+             ((uint32_t)p[3] << 24);
+    };
+    if (rd32(traw.data()) != kMagicSkippableSeek || rd32(traw.data() + 4) != tbl - 8)
+      throw std::runtime_error("zstd seek table: skippable-frame envelope mismatch");
     std::vector<MxzFrameHost> frames(nframes);
     uint64_t c_off = 0, d_off = 0;
     for (uint32_t i = 0; i < nframes; i++) {
       const uint8_t* e = traw.data() + 8 + (uint64_t)i * entry_sz;
-      uint32_t cs = (uint32_t)e[0] | ((uint32_t)e[1] << 8) | ((uint32_t)e[2] << 16) |
-                    ((uint32_t)e[3] << 24);
-      uint32_t ds = (uint32_t)e[4] | ((uint32_t)e[5] << 8) | ((uint32_t)e[6] << 16) |
-                    ((uint32_t)e[7] << 24);
+      uint32_t cs = rd32(e);
+      uint32_t ds = rd32(e + 4);
       frames[i] = {c_off, cs, d_off, ds};
       c_off += cs;
       d_off += ds;
     }
+    if (c_off != src_len - tbl)
+      throw std::runtime_error("zstd seek table does not cover the frames (" +
+                               std::to_string(c_off) + " != " + std::to_string(src_len - tbl) +
+                               ")");
     if (d_off > dst_cap) throw std::runtime_error("zstd decompress: dst too small");
     uint64_t batch = std::min<uint64_t>(nframes ? nframes : 1, 8192);
     std::lock_guard<std::mutex> zlk(zstd_mu_);
@@ -1031,6 +1126,9 @@ PYBIND11_MODULE(_core, m) {
       .def("sha256_chunk_leaves", &GpuEngine::sha256_chunk_leaves, py::arg("dev_ptr"),
            py::arg("size"), py::arg("chunk_size"))
       .def("sha256_multibuf", &GpuEngine::sha256_multibuf, py::arg("buffers"))
+      .def("sha256_canonical_device", &GpuEngine::sha256_canonical_device,
+           py::arg("src_ptr"), py::arg("size"))
+      .def("read_device", &GpuEngine::read_device, py::arg("src_ptr"), py::arg("size"))
       .def("push_part_from_device", &GpuEngine::push_part_from_device, py::arg("url"),
            py::arg("method"), py::arg("headers"), py::arg("src_ptr"), py::arg("size"))
       .def("tar_index", &GpuEngine::tar_index, py::arg("tar_ptr"), py::arg("tar_len"))
@@ -1063,6 +1161,23 @@ PYBIND11_MODULE(_core, m) {
         return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
       },
       py::arg("data"), py::arg("frame_raw") = (uint32_t)(128 << 10));
+  // CPU-testable canonical sha256 over host bytes — same OpenSSL EVP code
+  // the D2H canonical path runs (lets the GPU-less suite oracle it against
+  // hashlib)
+  m.def("sha256_host", [](py::bytes data) {
+    std::string s = data;
+    unsigned char md[32];
+    unsigned int mdlen = 0;
+    EVP_MD_CTX* ctx = EVP_MD_CTX_new();
+    if (!ctx || EVP_DigestInit_ex(ctx, EVP_sha256(), nullptr) != 1) {
+      if (ctx) EVP_MD_CTX_free(ctx);
+      throw std::runtime_error("EVP sha256 init failed");
+    }
+    EVP_DigestUpdate(ctx, s.data(), s.size());
+    EVP_DigestFinal_ex(ctx, md, &mdlen);
+    EVP_MD_CTX_free(ctx);
+    return py::bytes(reinterpret_cast<char*>(md), 32);
+  });
   m.def("zstd_decompress_cpu", [](py::bytes data) {
     std::string s = data;
     std::vector<uint8_t> out;

@@ -108,7 +108,13 @@ bool write_stream_to(const std::string& path, http::Request& req, std::string* e
       while (off < r) {
         ssize_t w = ::write(fd, small + off, static_cast<size_t>(r - off));
         if (w < 0 && errno == EINTR) continue;
-        if (w < 0) break;
+        if (w < 0) {
+          // counting unwritten bytes would let a short object pass the
+          // completeness check below and be renamed into place
+          ::close(fd);
+          ::unlink(tmp.c_str());
+          return false;
+        }
         off += w;
       }
       total += r;

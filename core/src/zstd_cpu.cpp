@@ -111,7 +111,10 @@ static bool skim_frame(const uint8_t* p, size_t len, uint64_t* csize, uint64_t* 
       dtot += bsize;
     } else if (btype == 2) {
       pos += bsize;
-      dtot = have_fcs ? dtot : dtot;  // compressed: regen unknown without decode
+      // compressed block: regenerated size is unknowable without decoding —
+      // without a Frame_Content_Size the frame walk cannot report d_size,
+      // so reject instead of returning a wrong size
+      if (!have_fcs) return false;
     } else {
       return false;
     }

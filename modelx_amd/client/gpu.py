@@ -111,10 +111,11 @@ class GpuClient:
                 cs = dg.algo_chunk_size(note.split(":", 1)[0]) or dg.DEFAULT_CHUNK_SIZE
                 expect = note
         if expect is None:
-            # canonical sha256 only: single sequential chain — GPU-hash via
-            # multibuf (1 chain); still verifies, just not at chunk rate
-            digests = self.engine.sha256_multibuf([(ptr, size)])
-            got = "sha256:" + digests[:32].hex()
+            # canonical sha256 only (no chunk annotation): one sequential
+            # chain — D2H through the pinned ring + SHA-NI on CPU, the fast
+            # path for a single chain (a GPU lane is ~4x slower)
+            digest = self.engine.sha256_canonical_device(ptr, size)
+            got = "sha256:" + digest.hex()
             if got != desc.digest:
                 raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
                                      f"GPU digest mismatch for {desc.name}: {got}")
@@ -125,16 +126,56 @@ class GpuClient:
             raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
                                  f"GPU chunk digest mismatch for {desc.name}: {got} != {expect}")
 
-    def _download_url(self, repository: str, desc: types.Descriptor) -> Tuple[str, Dict[str, str]]:
+    def _download_url(self, repository: str,
+                      desc: types.Descriptor) -> Optional[Tuple[str, Dict[str, str]]]:
+        """Presigned GET URL for a blob, or None when the registry runs
+        without --enable-redirect (the caller degrades to streaming the
+        bytes through the registry, pull.go:206-215 semantics)."""
         loc = self.remote.get_blob_location(repository, desc, "download")
         if loc is None:
-            raise er.ModelxError(er.ErrCode.UNSUPPORTED,
-                                 "registry has no presigned download location "
-                                 "(GPU pull needs --enable-redirect)")
+            return None
         parts = loc.properties.get("parts") or []
         if not parts:
             raise er.ModelxError(er.ErrCode.UNKNOWN, "empty location parts")
         return parts[0]["url"], _signed_headers(parts[0])
+
+    def _stream_via_registry(self, repository: str, desc: types.Descriptor,
+                             tensor) -> None:
+        """Fallback landing path for a redirect-less registry: stream the
+        blob through the registry API into HBM in bounded pinned pieces.
+        Slow (single stream, extra hop) but degrades instead of failing."""
+        import time
+
+        import torch
+
+        t0 = time.monotonic()
+        staging = torch.empty(self.slot_bytes, dtype=torch.uint8, pin_memory=True)
+        fill = 0
+        off = 0
+
+        def flush():
+            nonlocal fill, off
+            if fill:
+                tensor[off:off + fill].copy_(staging[:fill], non_blocking=False)
+                off += fill
+                fill = 0
+
+        for chunk in self.remote.get_blob_content(repository, desc.digest):
+            view = memoryview(chunk)
+            while view:
+                n = min(len(view), self.slot_bytes - fill)
+                staging[fill:fill + n] = torch.frombuffer(bytearray(view[:n]),
+                                                          dtype=torch.uint8)
+                fill += n
+                view = view[n:]
+                if fill == self.slot_bytes:
+                    flush()
+        flush()
+        if off != desc.size:
+            raise er.ModelxError(er.ErrCode.UNKNOWN,
+                                 f"registry stream truncated: {off} != {desc.size}")
+        self.last_stats.append({"phase": "pull-registry-stream", "bytes": off,
+                                "seconds": time.monotonic() - t0})
 
     # -------------------------------------------------------------- pull --
 
@@ -246,21 +287,28 @@ class GpuClient:
             # diagnostic: persist the (digest-verified) compressed bytes so a
             # decode failure is analyzable offline; then retry once — a
             # second identical failure is deterministic (bad stored frame),
-            # a pass means a device-side race to hunt
+            # a pass means a device-side race to hunt. Opt-in via
+            # MODELX_ZSTD_DUMP=<dir>: the blob can be gigabytes, so never
+            # write it into an arbitrary cwd by default.
+            import os as _os
             import sys
 
-            try:
-                os_dir = "gpurun_out"
-                import os as _os
-
-                _os.makedirs(os_dir, exist_ok=True)
-                path = f"{os_dir}/zstd_decode_fail_{desc.digest.split(':')[-1][:12]}.zst"
-                with open(path, "wb") as f:
-                    f.write(bytes(comp[: desc.size].cpu().numpy().tobytes()))
-                print(f"modelx: zstd decode failed ({e}); dumped {path}; retrying once",
-                      file=sys.stderr)
-            except Exception:
-                pass
+            dump_dir = _os.environ.get("MODELX_ZSTD_DUMP", "")
+            if dump_dir:
+                try:
+                    _os.makedirs(dump_dir, exist_ok=True)
+                    path = _os.path.join(
+                        dump_dir,
+                        f"zstd_decode_fail_{desc.digest.split(':')[-1][:12]}.zst")
+                    with open(path, "wb") as f:
+                        f.write(bytes(comp[: desc.size].cpu().numpy().tobytes()))
+                    print(f"modelx: zstd decode failed ({e}); dumped {path}; retrying once",
+                          file=sys.stderr)
+                except Exception:
+                    pass
+            else:
+                print(f"modelx: zstd decode failed ({e}); retrying once "
+                      "(set MODELX_ZSTD_DUMP=<dir> to keep the blob)", file=sys.stderr)
             got = self.engine.zstd_decompress_device(comp.data_ptr(), desc.size,
                                                      out.data_ptr(), out.numel())
         self.last_stats.append({"phase": "pull-zstd-decompress", "bytes": got,
@@ -297,7 +345,21 @@ class GpuClient:
             tensor = torch.empty(desc.size, dtype=torch.uint8, device=f"cuda:{self.device}")
         assert tensor.numel() >= desc.size
         planned = self._plan_url(plan_entry)
-        url, headers = planned if planned else self._download_url(repository, desc)
+        located = planned if planned else self._download_url(repository, desc)
+        if located is None:
+            # redirect-less registry: degrade to the streaming fallback
+            for attempt in range(2):
+                self._stream_via_registry(repository, desc, tensor)
+                if not verify:
+                    return tensor
+                try:
+                    self._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+                    return tensor
+                except er.ModelxError:
+                    if attempt:
+                        raise
+            return tensor
+        url, headers = located
 
         def expected_leaves():
             return (self._plan_leaves(plan_entry, desc)
@@ -549,16 +611,38 @@ class GpuClient:
                                 "seconds": time.monotonic() - t0})
         return root, leaves
 
+    PART_RETRIES = 3  # reference: pkg/client/extension_s3.go:133-148
+
     def _upload_small(self, repository: str, desc: types.Descriptor, data: bytes) -> None:
         loc = self.remote.get_blob_location(repository, desc, "upload")
         if loc is not None:
             import requests
 
             p = (loc.properties.get("parts") or [{}])[0]
-            requests.request(p.get("method") or "PUT", p["url"],
-                             headers=_signed_headers(p), data=data).raise_for_status()
+            for attempt in range(self.PART_RETRIES):
+                try:
+                    requests.request(p.get("method") or "PUT", p["url"],
+                                     headers=_signed_headers(p),
+                                     data=data).raise_for_status()
+                    return
+                except Exception:
+                    if attempt == self.PART_RETRIES - 1:
+                        raise
         else:
             self.remote.upload_blob_content(repository, desc, data)
+
+    def _push_part_retrying(self, part: dict, ptr: int, length: int) -> None:
+        """One presigned part from device memory, retried on transport
+        failure with a fresh connection (the engine drops broken sockets
+        from its pool on error)."""
+        for attempt in range(self.PART_RETRIES):
+            try:
+                self.engine.push_part_from_device(part["url"], part.get("method") or "PUT",
+                                                  _signed_headers(part), ptr, length)
+                return
+            except RuntimeError:
+                if attempt == self.PART_RETRIES - 1:
+                    raise
 
     def push_blob_from_device(self, repository: str, desc: types.Descriptor, ptr: int,
                               part_bytes: int = DEFAULT_PART_BYTES,
@@ -576,8 +660,21 @@ class GpuClient:
         extra = {"part-count": str(nparts), "multipart": "true"} if nparts > 1 else None
         loc = self.remote.get_blob_location(repository, desc, "upload", extra=extra)
         if loc is None:
-            raise er.ModelxError(er.ErrCode.UNSUPPORTED,
-                                 "registry has no presigned upload location")
+            # redirect-less registry: direct PUT through the registry
+            # (reference pushBlob fallback, push.go:196-207), streamed D2H
+            # in bounded slot-sized pieces
+            def body():
+                off = 0
+                while off < size:
+                    n = min(self.slot_bytes, size - off)
+                    yield self.engine.read_device(ptr + off, n)
+                    off += n
+
+            self.remote.upload_blob_content(repository, desc,
+                                            body() if size else b"")
+            self.last_stats.append({"phase": "push-registry-stream", "bytes": size,
+                                    "seconds": time.monotonic() - t0})
+            return
         parts = loc.properties.get("parts") or []
         ranges = []
         base = size // len(parts)
@@ -587,17 +684,13 @@ class GpuClient:
             ranges.append((off, ln))
             off += ln
         if len(parts) == 1:
-            p = parts[0]
-            self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
-                                              _signed_headers(p), ptr, size)
+            self._push_part_retrying(parts[0], ptr, size)
         else:
             from concurrent.futures import ThreadPoolExecutor
 
             def send(i):
-                p = parts[i]
                 o, ln = ranges[i]
-                self.engine.push_part_from_device(p["url"], p.get("method") or "PUT",
-                                                  _signed_headers(p), ptr + o, ln)
+                self._push_part_retrying(parts[i], ptr + o, ln)
 
             with ThreadPoolExecutor(max_workers=parallel) as pool:
                 list(pool.map(send, range(len(parts))))
@@ -608,16 +701,28 @@ class GpuClient:
                       tensors: Dict[str, "torch.Tensor"], config_yaml: str = "",
                       chunk_size: int = DEFAULT_GPU_CHUNK,
                       part_bytes: int = DEFAULT_PART_BYTES,
-                      compress: str = "") -> types.Manifest:
+                      compress: str = "", digest_mode: str = "chunked") -> types.Manifest:
         """Digest on GPU → presigned multipart upload → manifest PUT last.
         With ``compress="zstd"`` each tensor is compressed on-GPU into a
         seekable multi-frame zstd blob (core/hip/zstd.hip) first; the
         descriptor digest covers the stored (compressed) bytes and the
-        raw-digest/raw-size annotations carry the uncompressed identity."""
+        raw-digest/raw-size annotations carry the uncompressed identity.
+
+        ``digest_mode="sha256"`` makes the main descriptor digest the
+        wire-canonical plain sha256 (reference push.go:149-161 semantics —
+        a stock Go modelx client digest-verifies the pulled blob with zero
+        annotations consumed); the chunked digest then travels in the
+        chunk-digest annotation, exactly like the CPU push path. The
+        canonical chain streams D2H through the pinned ring onto the CPU's
+        SHA-NI units. Default ``"chunked"`` keeps the all-GPU digest as the
+        main digest (fastest; reference clients lose only verification)."""
         import torch
 
         if compress not in ("", "zstd"):
             raise er.ModelxError(er.ErrCode.UNSUPPORTED, f"unknown compression {compress!r}")
+        if digest_mode not in ("chunked", "sha256"):
+            raise er.ModelxError(er.ErrCode.UNSUPPORTED,
+                                 f"unknown digest_mode {digest_mode!r}")
         manifest = types.Manifest(media_type=types.MEDIA_TYPE_MODEL_MANIFEST_JSON)
         now = datetime.now(timezone.utc)
         # config blob (small, CPU)
@@ -656,9 +761,19 @@ class GpuClient:
                 push_ptr, push_size = comp_keep.data_ptr(), comp_size
             root, leaves = self.digest_device_blob_with_leaves(push_ptr, push_size,
                                                                chunk_size)
+            main_digest = root
+            if digest_mode == "sha256":
+                import time as _time
+
+                t0 = _time.monotonic()
+                canon = self.engine.sha256_canonical_device(push_ptr, push_size)
+                main_digest = "sha256:" + canon.hex()
+                self.last_stats.append({"phase": "push-canonical-digest",
+                                        "bytes": push_size,
+                                        "seconds": _time.monotonic() - t0})
             leaves_digest = dg.sha256_digest(leaves)
             desc = types.Descriptor(
-                name=name, media_type=media_type, digest=root, size=push_size,
+                name=name, media_type=media_type, digest=main_digest, size=push_size,
                 modified=now,
                 annotations={types.ANNOTATION_CHUNK_DIGEST: root,
                              types.ANNOTATION_CHUNK_SIZE: str(chunk_size),

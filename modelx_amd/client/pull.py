@@ -35,7 +35,6 @@ class Puller:
     def __init__(self, remote, concurrency: int = PULL_PUSH_CONCURRENCY):
         self.remote = remote
         self.concurrency = concurrency
-        self._plan_blobs: dict = {}
 
     def pull(self, repository: str, version: str, into_dir: str,
              blob_filter=None, quiet: Optional[bool] = None) -> types.Manifest:
@@ -49,56 +48,57 @@ class Puller:
             plan = None
         if plan and plan.get("manifest"):
             manifest = types.Manifest.from_dict(plan["manifest"])
-            self._plan_blobs = plan.get("blobs") or {}
+            plan_blobs = plan.get("blobs") or {}
         else:
             manifest = self.remote.get_manifest(repository, version)
-            self._plan_blobs = {}
+            plan_blobs = {}
         descs: List[types.Descriptor] = [manifest.config] + list(manifest.blobs)
         if blob_filter is not None:
             descs = [d for d in descs if blob_filter(d)]
-        try:
-            self.pull_blobs(repository, descs, into_dir, quiet=quiet)
-        finally:
-            self._plan_blobs = {}
+        self.pull_blobs(repository, descs, into_dir, quiet=quiet,
+                        plan_blobs=plan_blobs)
         return manifest
 
     def pull_blobs(self, repository: str, descs: List[types.Descriptor], into_dir: str,
-                   quiet: Optional[bool] = None) -> None:
+                   quiet: Optional[bool] = None, plan_blobs: Optional[dict] = None) -> None:
+        # the pull plan is threaded through as a parameter (not instance
+        # state) so concurrent pull() calls on one Puller can't cross plans
         with MultiBar("pull", self.concurrency, quiet=quiet) as mb:
             for desc in descs:
                 mb.go(desc.name, desc.size,
-                      lambda bar, d=desc: self._pull_one(repository, d, into_dir, bar))
+                      lambda bar, d=desc: self._pull_one(repository, d, into_dir, bar,
+                                                         plan_blobs))
             mb.wait()
 
     def _pull_one(self, repository: str, desc: types.Descriptor, into_dir: str,
-                  bar: Optional[Bar]) -> None:
+                  bar: Optional[Bar], plan_blobs: Optional[dict] = None) -> None:
         if desc.media_type in (types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ,
                                types.MEDIA_TYPE_MODEL_DIRECTORY_TAR):
-            self._pull_directory(repository, desc, into_dir, bar)
+            self._pull_directory(repository, desc, into_dir, bar, plan_blobs)
         elif desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
             if bar:
                 bar.set_status("sidecar", complete=True)
         elif desc.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
-            self._pull_zstd_file(repository, desc, into_dir, bar)
+            self._pull_zstd_file(repository, desc, into_dir, bar, plan_blobs)
         else:
-            self._pull_file(repository, desc, into_dir, bar)
+            self._pull_file(repository, desc, into_dir, bar, plan_blobs)
 
     # ------------------------------------------------------------- files --
 
     def _pull_file(self, repository: str, desc: types.Descriptor, into_dir: str,
-                   bar: Optional[Bar]) -> None:
+                   bar: Optional[Bar], plan_blobs: Optional[dict] = None) -> None:
         dest = os.path.join(into_dir, desc.name)
         # skip when the local file already matches (pull.go:115-124)
         if os.path.isfile(dest) and _verify_digest_of_file(dest, desc.digest):
             if bar:
                 bar.set_status("up to date", complete=True)
             return
-        self.pull_blob(repository, desc, dest, bar)
+        self.pull_blob(repository, desc, dest, bar, plan_blobs=plan_blobs)
         if desc.mode:
             os.chmod(dest, desc.mode & 0o7777)
 
     def _pull_zstd_file(self, repository: str, desc: types.Descriptor, into_dir: str,
-                        bar: Optional[Bar]) -> None:
+                        bar: Optional[Bar], plan_blobs: Optional[dict] = None) -> None:
         """CPU pull of a +zstd blob (GPU path: GpuClient.pull_zstd_blob_to_device).
         The stored bytes are verified by pull_blob against desc.digest; the
         decompressed output is verified against the raw-digest annotation."""
@@ -115,7 +115,7 @@ class Puller:
         os.makedirs(cache_dir, exist_ok=True)
         comp_path = os.path.join(cache_dir, desc.name + ".zst")
         if not (os.path.isfile(comp_path) and _verify_digest_of_file(comp_path, desc.digest)):
-            self.pull_blob(repository, desc, comp_path, bar)
+            self.pull_blob(repository, desc, comp_path, bar, plan_blobs=plan_blobs)
         with open(comp_path, "rb") as f:
             raw = _core.zstd_decompress_cpu(f.read())
         if raw_digest and not dg.verify_bytes(raw, raw_digest):
@@ -131,7 +131,7 @@ class Puller:
     # ------------------------------------------------------- directories --
 
     def _pull_directory(self, repository: str, desc: types.Descriptor, into_dir: str,
-                        bar: Optional[Bar]) -> None:
+                        bar: Optional[Bar], plan_blobs: Optional[dict] = None) -> None:
         from .helper import tgz, untgz
 
         compressed = desc.media_type == types.MEDIA_TYPE_MODEL_DIRECTORY_TARGZ
@@ -149,25 +149,28 @@ class Puller:
         tgz_path = os.path.join(cache_dir, desc.name + suffix)
         # cached two-phase: download then extract (pull.go:158-182)
         if not (os.path.isfile(tgz_path) and _verify_digest_of_file(tgz_path, desc.digest)):
-            self.pull_blob(repository, desc, tgz_path, bar)
+            self.pull_blob(repository, desc, tgz_path, bar, plan_blobs=plan_blobs)
         untgz(tgz_path, target, compressed=compressed)
 
     # -------------------------------------------------------------- blobs --
 
-    def _planned_location(self, desc: types.Descriptor) -> Optional[types.BlobLocation]:
-        entry = self._plan_blobs.get(desc.digest) if self._plan_blobs else None
+    @staticmethod
+    def _planned_location(desc: types.Descriptor,
+                          plan_blobs: Optional[dict]) -> Optional[types.BlobLocation]:
+        entry = plan_blobs.get(desc.digest) if plan_blobs else None
         loc = (entry or {}).get("location")
         if not loc:
             return None
         return types.BlobLocation.from_dict(loc)
 
     def pull_blob(self, repository: str, desc: types.Descriptor, dest_path: str,
-                  bar: Optional[Bar] = None, verify: bool = True) -> None:
+                  bar: Optional[Bar] = None, verify: bool = True,
+                  plan_blobs: Optional[dict] = None) -> None:
         """Presigned-location download with registry-stream fallback
         (pull.go:206-215), then digest verification (the reference never
         verifies after download — we do, and re-fetch once on mismatch)."""
         for attempt in range(2):
-            location = self._planned_location(desc) if attempt == 0 else None
+            location = self._planned_location(desc, plan_blobs) if attempt == 0 else None
             if location is None:
                 location = self.remote.get_blob_location(repository, desc, "download")
             if location is not None:
