@@ -5,6 +5,7 @@
 #include <atomic>
 #include <memory>
 #include <string>
+#include <vector>
 
 #include "modelx/http.hpp"
 #include "modelx/store.hpp"
@@ -12,12 +13,35 @@
 namespace modelx {
 namespace registry {
 
+// One RSA public key from a JWKS document (RFC 7517); n/e raw big-endian.
+struct JwksKey {
+  std::string kid;
+  std::string n;
+  std::string e;
+};
+
 struct AuthConfig {
   // Static bearer tokens (comma-separated via --auth-tokens); empty = open.
   std::vector<std::string> tokens;
   // HS256 JWT shared secret (offline OIDC-style verification); empty = off.
   std::string jwt_hs256_secret;
+  // Offline OIDC: RS256 ID tokens verified against these JWKS keys
+  // (--oidc-jwks FILE; reference helper.go:63-96 minus the egress-needing
+  // issuer discovery). Empty = off.
+  std::vector<JwksKey> jwks;
+  std::string oidc_issuer;    // checked against `iss` when non-empty
+  std::string oidc_audience;  // checked against `aud` when non-empty
+  bool enabled() const {
+    return !tokens.empty() || !jwt_hs256_secret.empty() || !jwks.empty();
+  }
 };
+
+bool load_jwks_file(const std::string& path, std::vector<JwksKey>* out, std::string* err);
+bool verify_jwt_hs256(const std::string& token, const std::string& secret,
+                      std::string* subject);
+bool verify_jwt_rs256(const std::string& token, const std::vector<JwksKey>& keys,
+                      const std::string& issuer, const std::string& audience,
+                      std::string* subject);
 
 // Prometheus-style counters (reference has no metrics endpoint — SURVEY.md
 // §5 observability gap; this is the /metrics the new framework adds).

@@ -49,6 +49,15 @@ int main(int argc, char** argv) {
         pos = comma + 1;
       }
     } else if (a == "--jwt-hs256-secret") auth.jwt_hs256_secret = next();
+    else if (a == "--oidc-jwks") {
+      std::string path = next();
+      std::string err;
+      if (!registry::load_jwks_file(path, &auth.jwks, &err)) {
+        fprintf(stderr, "modelxd: --oidc-jwks: %s\n", err.c_str());
+        return 1;
+      }
+    } else if (a == "--oidc-issuer") auth.oidc_issuer = next();
+    else if (a == "--oidc-audience") auth.oidc_audience = next();
     else if (a == "--tls-cert") tls.cert_file = next();
     else if (a == "--tls-key") tls.key_file = next();
     else if (a == "--help" || a == "-h") {
@@ -62,6 +71,9 @@ int main(int argc, char** argv) {
              "  --enable-redirect         hand out presigned S3 locations\n"
              "  --auth-tokens T1,T2       static bearer tokens\n"
              "  --jwt-hs256-secret S      verify HS256 JWTs offline\n"
+             "  --oidc-jwks FILE          verify RS256 ID tokens against a JWKS document\n"
+             "  --oidc-issuer ISS         require `iss` claim to equal ISS\n"
+             "  --oidc-audience AUD       require `aud` claim to contain AUD\n"
              "  --tls-cert F --tls-key F  serve HTTPS (reference --tls-*, server.go:37-43)\n");
       return 0;
     }

@@ -80,7 +80,7 @@ static bool valid_reference(const std::string& s) {
 }
 
 bool Registry::authorize(http::Request& req, http::ResponseWriter& w) {
-  if (auth_.tokens.empty() && auth_.jwt_hs256_secret.empty()) return true;
+  if (!auth_.enabled()) return true;
   std::string token;
   auto it = req.headers.find("Authorization");
   if (it != req.headers.end() && it->second.rfind("Bearer ", 0) == 0)
@@ -97,10 +97,14 @@ bool Registry::authorize(http::Request& req, http::ResponseWriter& w) {
   for (auto& t : auth_.tokens)
     if (t == token) return true;
   if (!auth_.jwt_hs256_secret.empty()) {
-    extern bool verify_jwt_hs256(const std::string& token, const std::string& secret,
-                                 std::string* subject);
     std::string subject;
     if (verify_jwt_hs256(token, auth_.jwt_hs256_secret, &subject)) return true;
+  }
+  if (!auth_.jwks.empty()) {
+    std::string subject;
+    if (verify_jwt_rs256(token, auth_.jwks, auth_.oidc_issuer, auth_.oidc_audience,
+                         &subject))
+      return true;
   }
   response_error(w, wire::ErrorInfo{401, "UNAUTHORIZED", "invalid token", ""});
   return false;
