@@ -42,6 +42,15 @@ def parse_args():
     p.add_argument("--store", default="", help="s3d store root (default: tmpfs)")
     p.add_argument("--cpu-smoke", action="store_true",
                    help="CPU-only plumbing run (BASELINE config 1, 10 MiB)")
+    p.add_argument("--mode", choices=["auto", "fanout", "independent"], default="auto",
+                   help="multi-GPU step shape: 'fanout' = one shared store, each rank "
+                        "pushes 1/N of a fixed-size index, every shard RCCL-broadcast "
+                        "over xGMI from its owner so ALL GPUs land the full index, "
+                        "digest-verified per GPU (BASELINE config 4 semantics — the "
+                        "mode SCALE runs); 'independent' = each rank its own stack "
+                        "and blob (no collectives). auto = fanout when world>1.")
+    p.add_argument("--chunk-mib", type=int, default=256,
+                   help="fan-out broadcast pipeline chunk (MiB)")
     return p.parse_args()
 
 
@@ -164,18 +173,27 @@ def main():
     if not cpu_mode:
         torch.cuda.set_device(device)
 
-    # --- per-rank S3 + registry stack -------------------------------------
-    # Each rank runs its own modelxd + s3d shard (object stores scale
-    # horizontally; the measured entity is the per-GPU client data plane).
+    fanout = world > 1 and args.mode in ("auto", "fanout")  # needs collectives
+
+    # --- S3 + registry stack ----------------------------------------------
+    # fanout mode: ONE shared modelxd + s3d for the job (rank 0 starts them)
+    # — the pull path is one S3 fetch per shard fanned out over xGMI, so the
+    # store sees index-size bytes per step regardless of N. independent
+    # mode: each rank its own stack (object stores scale horizontally; the
+    # measured entity is the per-GPU client data plane).
     from util_servers import MODELXD, S3D, ServerProc, wait_http
 
     master_port = int(os.environ.get("MASTER_PORT", "29500"))
-    s3_port = master_port + 1371 + 2 * rank
-    mdx_port = master_port + 1372 + 2 * rank
-    store_root = pick_store_root(args) + f"-r{rank}"
+    if fanout:
+        s3_port = master_port + 1371
+        mdx_port = master_port + 1372
+        store_root = pick_store_root(args) + "-shared"
+    else:
+        s3_port = master_port + 1371 + 2 * rank
+        mdx_port = master_port + 1372 + 2 * rank
+        store_root = pick_store_root(args) + f"-r{rank}"
     procs = []
-    shutil.rmtree(store_root, ignore_errors=True)
-    os.makedirs(os.path.join(store_root, "modelx"), exist_ok=True)
+    owns_stack = (not fanout) or rank == 0
     if rank == 0 and not (os.path.exists(S3D) and os.path.exists(MODELXD)):
         import subprocess
 
@@ -184,18 +202,24 @@ def main():
         import torch.distributed as dist
 
         dist.barrier()  # wait for a possible rank-0 build
-    procs.append(ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root", store_root,
-                             "--access-key", "modelx", "--secret-key", "modelx123"], s3_port))
-    wait_http(s3_port)
-    procs.append(ServerProc(
-        [MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
-         f"http://127.0.0.1:{s3_port}", "--s3-bucket", "modelx", "--s3-access-key",
-         "modelx", "--s3-secret-key", "modelx123", "--enable-redirect"], mdx_port))
-    wait_http(mdx_port)
+    if owns_stack:
+        shutil.rmtree(store_root, ignore_errors=True)
+        os.makedirs(os.path.join(store_root, "modelx"), exist_ok=True)
+        procs.append(ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root",
+                                 store_root, "--access-key", "modelx",
+                                 "--secret-key", "modelx123"], s3_port))
+        wait_http(s3_port)
+        procs.append(ServerProc(
+            [MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
+             f"http://127.0.0.1:{s3_port}", "--s3-bucket", "modelx", "--s3-access-key",
+             "modelx", "--s3-secret-key", "modelx123", "--enable-redirect"], mdx_port))
+        wait_http(mdx_port)
     if distributed:
         import torch.distributed as dist
 
         dist.barrier()
+        if not owns_stack:
+            wait_http(mdx_port)
 
     if cpu_mode:
         g = _CpuBenchClient(f"http://127.0.0.1:{mdx_port}")
@@ -206,9 +230,19 @@ def main():
                       slot_bytes=args.slot_mib << 20, num_conns=args.conns)
 
     blob_bytes = int(args.blob_gib * (1 << 30))
-    src = torch.empty(blob_bytes, dtype=torch.uint8,
+    # fanout: the index size is FIXED at blob_gib as N grows — each rank
+    # pushes a 1/N shard, the fan-out lands the FULL index on every GPU
+    # (per-GPU landed bytes constant = weak scaling; the aggregate delivered
+    # bandwidth is what xGMI multiplies)
+    if fanout:
+        base = blob_bytes // world
+        shard_bytes = blob_bytes - base * (world - 1) if rank == world - 1 else base
+        repo = "bench/fanout"
+    else:
+        shard_bytes = blob_bytes
+        repo = f"bench/rank{rank}"
+    src = torch.empty(shard_bytes, dtype=torch.uint8,
                       device="cpu" if cpu_mode else f"cuda:{device}")
-    repo = f"bench/rank{rank}"
 
     substeps = []
 
@@ -219,14 +253,38 @@ def main():
         if not cpu_mode:
             torch.cuda.synchronize(device)
         t1 = time.monotonic()
-        g.push_from_gpu(repo, f"s{step_idx}", {"blob.bin": src},
-                        part_bytes=args.part_mib << 20)
-        t2 = time.monotonic()
-        g.pull_to_gpu(repo, f"s{step_idx}", verify=True)  # GPU digest verify
-        t3 = time.monotonic()
-        # drop this step's objects so tmpfs doesn't fill across steps
-        g.remote.delete_index(repo)
-        t4 = time.monotonic()
+        if fanout:
+            import torch.distributed as dist
+
+            from modelx_amd.client.fanout import fanout_pull_broadcast
+
+            g.push_from_gpu(repo, f"s{step_idx}-r{rank}", {"shard.bin": src},
+                            part_bytes=args.part_mib << 20)
+            dist.barrier()  # every shard published before the fan-out
+            t2 = time.monotonic()
+            landed = 0
+            for r in range(world):
+                out = fanout_pull_broadcast(dist, g, repo, f"s{step_idx}-r{r}",
+                                            device, chunk=args.chunk_mib << 20,
+                                            src_rank=r, verify=True)
+                landed += sum(t.numel() for t in out.values())
+                del out
+            assert landed == blob_bytes, (landed, blob_bytes)
+            t3 = time.monotonic()
+            dist.barrier()
+            if rank == 0:
+                g.remote.delete_index(repo)
+            dist.barrier()  # delete done before the next step's push
+            t4 = time.monotonic()
+        else:
+            g.push_from_gpu(repo, f"s{step_idx}", {"blob.bin": src},
+                            part_bytes=args.part_mib << 20)
+            t2 = time.monotonic()
+            g.pull_to_gpu(repo, f"s{step_idx}", verify=True)  # GPU digest verify
+            t3 = time.monotonic()
+            # drop this step's objects so tmpfs doesn't fill across steps
+            g.remote.delete_index(repo)
+            t4 = time.monotonic()
         substeps.append((t1 - t0, t2 - t1, t3 - t2, t4 - t3))
 
     def barrier_sync():
@@ -277,7 +335,17 @@ def main():
         p.stop()
     shutil.rmtree(store_root, ignore_errors=True)
     if rank == 0:
-        moved_gib = 2.0 * args.blob_gib * args.steps * world  # push + pull, all ranks
+        if fanout:
+            # per step: HBM→S3 push of the index (blob_gib aggregate across
+            # ranks) + S3/xGMI→HBM landing of the FULL index on every GPU
+            # (world × blob_gib, each digest-verified at its destination)
+            moved_gib = (1.0 + world) * args.blob_gib * args.steps
+            model = f"config4-{args.blob_gib:g}GiB-index-sharded{world}"
+            par = f"fanout{world}-rccl-xgmi-broadcast"
+        else:
+            moved_gib = 2.0 * args.blob_gib * args.steps * world  # push+pull, all ranks
+            model = f"synthetic-{args.blob_gib:g}GiB-blob"
+            par = f"dp{world}-presigned-s3"
         print(json.dumps({
             "metric": "push+pull GiB/s end-to-end (S3->HBM, digest-verified), 1/2/4/8 MI355X",
             "value": round(moved_gib / elapsed, 3),
@@ -291,9 +359,9 @@ def main():
             "vs_baseline": None,  # reference publishes no numbers
             "dtype": "uint8",
             "data": "synthetic",
-            "config": {"model": f"synthetic-{args.blob_gib:g}GiB-blob",
+            "config": {"model": model,
                        "global_batch": world, "seq_len": 0,
-                       "parallelism": f"dp{world}-presigned-s3"}}), flush=True)
+                       "parallelism": par}}), flush=True)
     if distributed:
         import torch.distributed as dist
 

@@ -791,6 +791,15 @@ MX_HD static inline i64 decode_block(DecCtx* c, u32* rep, const u8* src, u64 src
 
   u64 lit_used = 0;
   for (u32 s = 0; s < nseq; s++) {
+    // bound every FSE state before it indexes a table: a corrupt table
+    // description can break the base+nbits invariant, and one
+    // slightly-out-of-range state reads garbage base[] that catapults the
+    // next state far outside the struct (found by ASAN replay of the
+    // decoder mutation fuzz — intermittent SEGV). Applies to the GPU
+    // decoder identically (device OOB reads crash the box).
+    if (stLL >= (1u << c->ll.log) || stOF >= (1u << c->of.log) ||
+        stML >= (1u << c->ml.log))
+      return MXZ_ERR_SEQUENCES;
     u32 ofCode = c->of.symbol[stOF];
     u32 mlCode = c->ml.symbol[stML];
     u32 llCode = c->ll.symbol[stLL];

@@ -84,31 +84,58 @@ def broadcast_blob_pipelined(dist, tensor, size: int, src_rank: int,
 
 def fanout_pull_broadcast(dist, gpu_client, repository: str, version: str,
                           device: int, chunk: int = DEFAULT_PIPELINE_CHUNK,
-                          src_rank: int = 0, verify: bool = True):
+                          src_rank: int = 0, verify: bool = True,
+                          skip_sidecars: bool = True):
     """All-replicate pull: rank `src_rank` fetches from S3, every rank ends
     with all blobs in its HBM (BASELINE config 4 replicate mode). Every rank
     verifies its own copy on-GPU — corruption anywhere in S3, host ring,
-    PCIe or xGMI is caught at the destination."""
+    PCIe or xGMI is caught at the destination.
+
+    Accepts any client with the GpuClient surface; a client without an
+    ``engine`` (the bench's CPU dry-run stand-in) runs the SAME chunk
+    schedule and collectives on CPU tensors over gloo — the choreography is
+    identical, only the transfer/verify callables differ."""
     import torch
 
+    is_gpu = hasattr(gpu_client, "engine")
+    dev = f"cuda:{device}" if is_gpu else "cpu"
     rank = dist.get_rank()
     manifest = gpu_client.remote.get_manifest(repository, version)
     out = {}
     for desc in manifest.blobs:
         if desc.size == 0:
             continue
-        tensor = torch.empty(desc.size, dtype=torch.uint8, device=f"cuda:{device}")
+        if skip_sidecars and desc.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
+            continue
+        tensor = torch.empty(desc.size, dtype=torch.uint8, device=dev)
         fetch = None
         if rank == src_rank:
-            url, headers = gpu_client._download_url(repository, desc)
+            if is_gpu:
+                located = gpu_client._download_url(repository, desc)
+                if located is None:
+                    raise RuntimeError("fan-out needs a presigned location "
+                                       "(registry without --enable-redirect)")
+                url, headers = located
 
-            def fetch(off, ln, _url=url, _h=headers, _t=tensor):
-                gpu_client.engine.pull_to_device(_url, _h, ln, _t.data_ptr() + off,
-                                                 gpu_client.num_conns, off)
-
+                def fetch(off, ln, _url=url, _h=headers, _t=tensor):
+                    gpu_client.engine.pull_to_device(_url, _h, ln, _t.data_ptr() + off,
+                                                     gpu_client.num_conns, off)
+            else:
+                # CPU dry-run: materialize the blob before the broadcasts
+                buf = b"".join(gpu_client.remote.get_blob_content(repository, desc.digest))
+                tensor.copy_(torch.frombuffer(bytearray(buf), dtype=torch.uint8))
         broadcast_blob_pipelined(dist, tensor, desc.size, src_rank, fetch, chunk)
         if verify:
-            gpu_client._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+            if is_gpu:
+                gpu_client._verify_device_digest(tensor.data_ptr(), desc.size, desc)
+            else:
+                from ..wire import digest as dg
+
+                raw = tensor.numpy().tobytes()
+                target = desc.digest or desc.annotations.get(
+                    types.ANNOTATION_CHUNK_DIGEST, "")
+                if target and not dg.verify_bytes(raw, target):
+                    raise RuntimeError(f"fan-out digest mismatch for {desc.name}")
         out[desc.name] = tensor
     return out
 
