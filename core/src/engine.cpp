@@ -925,6 +925,172 @@ class GpuEngine {
     return d_off;
   }
 
+  // Batched decode of MANY seekable blobs in one call. The single-blob
+  // path costs ~5 stream round-trips per blob (footer, table, batch sync,
+  // rc check) serialized under zstd_mu_ — measured 4.1 GiB/s effective on
+  // config-5's 64 MiB blobs against a 240-540 GiB/s kernel. Here the
+  // footer/table parses collapse to two syncs for the whole set, decode
+  // launches spread across the engine streams, and ONE final sync checks
+  // every frame's rc. items = [(src_ptr, src_len, dst_ptr, dst_cap)];
+  // returns decompressed sizes.
+  std::vector<uint64_t> zstd_decompress_many(
+      const std::vector<std::tuple<uintptr_t, uint64_t, uintptr_t, uint64_t>>& items) {
+    HIP_CHECK(hipSetDevice(device_));
+    using namespace modelx::zstd;
+    size_t n = items.size();
+    std::vector<uint64_t> out(n, 0);
+    if (!n) return out;
+    py::gil_scoped_release release;
+    std::lock_guard<std::mutex> zlk(zstd_mu_);
+    auto rd32 = [](const uint8_t* p) {
+      return (uint32_t)p[0] | ((uint32_t)p[1] << 8) | ((uint32_t)p[2] << 16) |
+             ((uint32_t)p[3] << 24);
+    };
+    // phase 1: all footers, one sync
+    std::vector<uint8_t> foots(n * 17);
+    for (size_t i = 0; i < n; i++) {
+      auto [src, len, dst, cap] = items[i];
+      (void)dst;
+      (void)cap;
+      if (len < 17) throw std::runtime_error("zstd blob too small");
+      HIP_CHECK(hipMemcpyAsync(foots.data() + i * 17, reinterpret_cast<char*>(src) + len - 17,
+                               17, hipMemcpyDeviceToHost, hash_stream_));
+    }
+    HIP_CHECK(hipStreamSynchronize(hash_stream_));
+    std::vector<uint64_t> tbls(n), tbl_off(n);
+    std::vector<uint32_t> nframes(n);
+    std::vector<uint8_t> entry_sz(n);
+    uint64_t total_tbl = 0;
+    for (size_t i = 0; i < n; i++) {
+      const uint8_t* foot = foots.data() + i * 17;
+      if (rd32(foot + 13) != kSeekTableMagic)
+        throw std::runtime_error("zstd blob has no seek table (footer magic mismatch)");
+      nframes[i] = rd32(foot + 8);
+      entry_sz[i] = (foot[12] & 0x80) ? 12 : 8;
+      tbls[i] = 8 + (uint64_t)nframes[i] * entry_sz[i] + 9;
+      if (tbls[i] > std::get<1>(items[i]))
+        throw std::runtime_error("zstd seek table larger than blob");
+      tbl_off[i] = total_tbl;
+      total_tbl += tbls[i];
+    }
+    // phase 2: all tables, one sync
+    std::vector<uint8_t> traw(total_tbl);
+    for (size_t i = 0; i < n; i++) {
+      auto [src, len, dst, cap] = items[i];
+      (void)dst;
+      (void)cap;
+      HIP_CHECK(hipMemcpyAsync(traw.data() + tbl_off[i],
+                               reinterpret_cast<char*>(src) + len - tbls[i], tbls[i],
+                               hipMemcpyDeviceToHost, hash_stream_));
+    }
+    HIP_CHECK(hipStreamSynchronize(hash_stream_));
+    std::vector<MxzFrameHost> frames;
+    std::vector<std::pair<size_t, size_t>> item_span(n);  // [first, count) into frames
+    uint64_t max_chunk = 1;
+    for (size_t i = 0; i < n; i++) {
+      const uint8_t* t = traw.data() + tbl_off[i];
+      if (rd32(t) != kMagicSkippableSeek || rd32(t + 4) != tbls[i] - 8)
+        throw std::runtime_error("zstd seek table: skippable-frame envelope mismatch");
+      uint64_t c_off = 0, d_off = 0;
+      size_t first = frames.size();
+      for (uint32_t f = 0; f < nframes[i]; f++) {
+        const uint8_t* e = t + 8 + (uint64_t)f * entry_sz[i];
+        uint32_t cs = rd32(e), ds = rd32(e + 4);
+        frames.push_back({c_off, cs, d_off, ds});
+        c_off += cs;
+        d_off += ds;
+      }
+      if (c_off != std::get<1>(items[i]) - tbls[i])
+        throw std::runtime_error("zstd seek table does not cover the frames");
+      if (d_off > std::get<3>(items[i]))
+        throw std::runtime_error("zstd decompress: dst too small");
+      item_span[i] = {first, frames.size() - first};
+      out[i] = d_off;
+      max_chunk = std::max<uint64_t>(max_chunk,
+                                     std::min<uint64_t>(nframes[i] ? nframes[i] : 1, 8192));
+    }
+    if (frames.empty()) return out;
+    // phase 3: one frame-array H2D, decode launches across streams
+    size_t ns = streams_.size();
+    void* dframes = zstd_scratch(4, frames.size() * sizeof(MxzFrameHost));
+    int64_t* drc = reinterpret_cast<int64_t*>(
+        zstd_scratch(7, frames.size() * sizeof(int64_t)));
+    void* dlit = zstd_scratch(5, ns * max_chunk * (uint64_t)kBlockMax);
+    HIP_CHECK(hipMemcpyAsync(dframes, frames.data(), frames.size() * sizeof(MxzFrameHost),
+                             hipMemcpyHostToDevice, hash_stream_));
+    hipEvent_t staged;
+    HIP_CHECK(hipEventCreateWithFlags(&staged, hipEventDisableTiming));
+    HIP_CHECK(hipEventRecord(staged, hash_stream_));
+    for (size_t s = 0; s < ns; s++) HIP_CHECK(hipStreamWaitEvent(streams_[s], staged, 0));
+    size_t rr = 0;
+    for (size_t i = 0; i < n; i++) {
+      auto [src, len, dst, cap] = items[i];
+      (void)len;
+      (void)cap;
+      auto [first, count] = item_span[i];
+      for (size_t off = 0; off < count; off += max_chunk) {
+        uint32_t nb = (uint32_t)std::min<uint64_t>(max_chunk, count - off);
+        size_t sidx = rr++ % ns;
+        HIP_CHECK(modelx_zstd_decompress_frames(
+            reinterpret_cast<void*>(src),
+            static_cast<MxzFrameHost*>(dframes) + first + off, nb,
+            reinterpret_cast<void*>(dst),
+            static_cast<char*>(dlit) + sidx * max_chunk * (uint64_t)kBlockMax,
+            drc + first + off, zstd_flags(), streams_[sidx]));
+      }
+    }
+    for (auto& st : streams_) HIP_CHECK(hipStreamSynchronize(st));
+    hipEventDestroy(staged);
+    std::vector<int64_t> hrc(frames.size());
+    HIP_CHECK(hipMemcpy(hrc.data(), drc, frames.size() * sizeof(int64_t),
+                        hipMemcpyDeviceToHost));
+    for (size_t f = 0; f < hrc.size(); f++)
+      if (hrc[f] != 0)
+        throw std::runtime_error("zstd decompress kernel failed: frame " + std::to_string(f) +
+                                 " rc=" + std::to_string(hrc[f]));
+    return out;
+  }
+
+  // Batched chunk-leaf digests of many device buffers: all launches on
+  // hash_stream_, ONE sync, ONE D2H — the per-blob sha256_chunk_leaves
+  // round trips were the other orchestration term on many-small-blob
+  // indexes. items = [(ptr, size, chunk_size)].
+  std::vector<py::bytes> sha256_chunk_leaves_many(
+      const std::vector<std::tuple<uintptr_t, uint64_t, uint64_t>>& items) {
+    HIP_CHECK(hipSetDevice(device_));
+    size_t n = items.size();
+    std::vector<py::bytes> res;
+    if (!n) return res;
+    std::vector<uint32_t> nchunks(n);
+    std::vector<uint64_t> off(n);
+    uint64_t total = 0;
+    for (size_t i = 0; i < n; i++) {
+      auto [ptr, size, cs] = items[i];
+      (void)ptr;
+      nchunks[i] = size ? (uint32_t)((size + cs - 1) / cs) : 1;
+      off[i] = total;
+      total += nchunks[i];
+    }
+    std::string host;
+    {
+      py::gil_scoped_release release;
+      std::lock_guard<std::mutex> zlk(zstd_mu_);
+      void* dleaves = zstd_scratch(8, total * 32);
+      for (size_t i = 0; i < n; i++) {
+        auto [ptr, size, cs] = items[i];
+        HIP_CHECK(modelx_sha256_chunk_leaves(reinterpret_cast<void*>(ptr), size, cs,
+                                             static_cast<char*>(dleaves) + off[i] * 32,
+                                             nchunks[i], hash_stream_));
+      }
+      HIP_CHECK(hipStreamSynchronize(hash_stream_));
+      host.resize(total * 32);
+      HIP_CHECK(hipMemcpy(&host[0], dleaves, host.size(), hipMemcpyDeviceToHost));
+    }
+    for (size_t i = 0; i < n; i++)
+      res.emplace_back(host.data() + off[i] * 32, (size_t)nchunks[i] * 32);
+    return res;
+  }
+
   // Upper bound for zstd_compress_device output.
   static uint64_t zstd_compress_bound(uint64_t size, uint32_t frame_raw) {
     using namespace modelx::zstd;
@@ -1083,8 +1249,8 @@ class GpuEngine {
   std::mutex mu_;
   std::condition_variable cv_free_, cv_pending_;
   std::mutex zstd_mu_;
-  void* zs_ptr_[7] = {};
-  size_t zs_size_[7] = {};
+  void* zs_ptr_[10] = {};
+  size_t zs_size_[10] = {};
   ConnPool conn_pool_;
   std::mutex dedup_mu_;
   void* dedup_table_ = nullptr;
@@ -1138,6 +1304,9 @@ PYBIND11_MODULE(_core, m) {
            py::arg("chunk_size"), py::arg("total"))
       .def("dedup_pull", &GpuEngine::dedup_pull, py::arg("expect"), py::arg("dst_ptr"),
            py::arg("chunk_size"), py::arg("total"))
+      .def("zstd_decompress_many", &GpuEngine::zstd_decompress_many, py::arg("items"))
+      .def("sha256_chunk_leaves_many", &GpuEngine::sha256_chunk_leaves_many,
+           py::arg("items"))
       .def("zstd_compress_device", &GpuEngine::zstd_compress_device, py::arg("src_ptr"),
            py::arg("size"), py::arg("frame_raw"), py::arg("dst_ptr"), py::arg("dst_cap"))
       .def("zstd_decompress_device", &GpuEngine::zstd_decompress_device, py::arg("src_ptr"),

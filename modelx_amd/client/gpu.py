@@ -501,12 +501,107 @@ class GpuClient:
                                 "seconds": time.monotonic() - t0})
         return tensor
 
+    def _pull_zstd_batched(self, repository: str, jobs, verify: bool,
+                           parallel: int) -> Dict[object, "torch.Tensor"]:
+        """Many +zstd blobs in one orchestration pass: compressed bytes land
+        (and stream-hash-verify) per blob, then ONE batched decode call and
+        ONE batched leaf-digest call cover the whole set — the per-blob
+        footer/table/sync round trips were the measured config-5 limiter
+        (profiles/bench_shapes.md: 4.1 GiB/s effective vs 240+ GiB/s
+        kernel). jobs = [(key, desc, plan_entry)]; returns {key: tensor}.
+
+        Waves are capped at ~64 GiB of raw output so compressed+raw
+        residency stays bounded on very large sets."""
+        import time
+
+        import torch
+
+        wave_cap = 64 << 30
+        if len(jobs) > 1:
+            total_raw = sum(int(d.annotations.get(types.ANNOTATION_RAW_SIZE, 0) or 0)
+                            for _, d, _ in jobs)
+            if total_raw > wave_cap:
+                out: Dict[object, "torch.Tensor"] = {}
+                wave, acc = [], 0
+                for job in jobs:
+                    raw = int(job[1].annotations.get(types.ANNOTATION_RAW_SIZE, 0) or 0)
+                    if wave and acc + raw > wave_cap:
+                        out.update(self._pull_zstd_batched(repository, wave, verify,
+                                                           parallel))
+                        wave, acc = [], 0
+                    wave.append(job)
+                    acc += raw
+                if wave:
+                    out.update(self._pull_zstd_batched(repository, wave, verify,
+                                                       parallel))
+                return out
+
+        def one(job):
+            key, desc, entry = job
+            return key, self.pull_blob_to_device(repository, desc, verify=verify,
+                                                 plan_entry=entry)
+
+        if parallel > 1 and len(jobs) > 1:
+            from concurrent.futures import ThreadPoolExecutor
+
+            with ThreadPoolExecutor(max_workers=parallel) as ex:
+                comp = dict(ex.map(one, jobs))
+        else:
+            comp = dict(one(j) for j in jobs)
+        outs: Dict[object, "torch.Tensor"] = {}
+        items = []
+        metas = []
+        for key, desc, _ in jobs:
+            raw_size = int(desc.annotations.get(types.ANNOTATION_RAW_SIZE, 0) or 0)
+            if not raw_size:
+                raise er.ModelxError(er.ErrCode.UNSUPPORTED,
+                                     f"+zstd blob {desc.name} lacks the raw-size annotation")
+            o = torch.empty(max(raw_size, 1), dtype=torch.uint8,
+                            device=f"cuda:{self.device}")
+            items.append((comp[key].data_ptr(), desc.size, o.data_ptr(), o.numel()))
+            outs[key] = o
+            metas.append((key, desc, raw_size))
+        t0 = time.monotonic()
+        sizes = self.engine.zstd_decompress_many(items)
+        self.last_stats.append({"phase": "pull-zstd-decompress-batched",
+                                "bytes": sum(sizes), "blobs": len(items),
+                                "seconds": time.monotonic() - t0})
+        for (key, desc, raw_size), got in zip(metas, sizes):
+            if got != raw_size:
+                raise er.ModelxError(er.ErrCode.DIGEST_INVALID,
+                                     f"zstd size mismatch for {desc.name}: "
+                                     f"{got} != {raw_size}")
+            outs[key] = outs[key][:got]
+        if verify:
+            vitems, vmeta = [], []
+            for key, desc, raw_size in metas:
+                rd = desc.annotations.get(types.ANNOTATION_RAW_DIGEST, "")
+                if not rd:
+                    continue
+                cs = dg.algo_chunk_size(rd.split(":", 1)[0]) or DEFAULT_GPU_CHUNK
+                o = outs[key]
+                vitems.append((o.data_ptr(), o.numel(), cs))
+                vmeta.append((desc, cs, o.numel(), rd))
+            if vitems:
+                t0 = time.monotonic()
+                leaves_list = self.engine.sha256_chunk_leaves_many(vitems)
+                for (desc, cs, sz, rd), leaves in zip(vmeta, leaves_list):
+                    if dg.root_from_leaf_bytes(leaves, cs, sz) != rd:
+                        raise er.ModelxError(
+                            er.ErrCode.DIGEST_INVALID,
+                            f"uncompressed digest mismatch for {desc.name}")
+                self.last_stats.append({"phase": "pull-zstd-raw-verify-batched",
+                                        "bytes": sum(v[1] for v in vitems),
+                                        "seconds": time.monotonic() - t0})
+        return outs
+
     def pull_to_gpu(self, repository: str, version: str = "",
                     verify: bool = True, parallel: int = 1) -> Dict[str, "torch.Tensor"]:
         """Pull every file blob of a manifest into HBM. Directory (tar.gz)
         blobs are landed as raw archive bytes under their blob name.
         ``parallel`` > 1 pulls blobs concurrently (reentrant engine; right
-        for many-shard manifests where per-blob latency would stack)."""
+        for many-shard manifests where per-blob latency would stack).
+        Multiple +zstd blobs decode through the batched engine call."""
         # one round trip for manifest + presigns + leaves when the server
         # supports pull plans (measured: the per-blob control plane
         # dominated many-small-blob indexes — docs/roadmap.md)
@@ -519,6 +614,16 @@ class GpuClient:
             plan_blobs = {}
         descs = [d for d in manifest.blobs
                  if d.size and d.media_type != types.MEDIA_TYPE_MODEL_LEAVES]
+        zstd_descs = [d for d in descs
+                      if d.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD]
+        plain_descs = [d for d in descs
+                       if d.media_type != types.MEDIA_TYPE_MODEL_FILE_ZSTD]
+
+        out: Dict[str, "torch.Tensor"] = {}
+        if len(zstd_descs) > 1:
+            jobs = [(d.name, d, plan_blobs.get(d.digest)) for d in zstd_descs]
+            out.update(self._pull_zstd_batched(repository, jobs, verify, parallel))
+            zstd_descs = []
 
         def one(desc):
             entry = plan_blobs.get(desc.digest)
@@ -529,12 +634,15 @@ class GpuClient:
             return desc.name, self.pull_blob_to_device(repository, desc, verify=verify,
                                                        plan_entry=entry)
 
-        if parallel > 1 and len(descs) > 1:
+        rest = plain_descs + zstd_descs
+        if parallel > 1 and len(rest) > 1:
             from concurrent.futures import ThreadPoolExecutor
 
             with ThreadPoolExecutor(max_workers=parallel) as ex:
-                return dict(ex.map(one, descs))
-        return dict(one(d) for d in descs)
+                out.update(dict(ex.map(one, rest)))
+        else:
+            out.update(dict(one(d) for d in rest))
+        return out
 
     def pull_many(self, repository: str, versions, parallel: int = 6,
                   verify: bool = True) -> Dict[str, Dict[str, "torch.Tensor"]]:
@@ -542,14 +650,51 @@ class GpuClient:
         is reentrant (per-call range state, shared pinned-slot pool), so
         many small blobs overlap their HTTP round-trips — the limiter for
         mixed indexes (BASELINE config 5) is per-blob latency, not
-        bandwidth."""
+        bandwidth. All +zstd blobs ACROSS the versions decode through one
+        batched engine call (config 5 stores one small blob per version, so
+        per-version batching alone would never engage)."""
         from concurrent.futures import ThreadPoolExecutor
 
-        def one(v):
-            return v, self.pull_to_gpu(repository, v, verify=verify)
+        def info(v):
+            plan = self.remote.get_pull_plan(repository, v)
+            if plan and plan.get("manifest"):
+                return v, types.Manifest.from_dict(plan["manifest"]), plan.get("blobs") or {}
+            return v, self.remote.get_manifest(repository, v), {}
 
         with ThreadPoolExecutor(max_workers=parallel) as ex:
-            return dict(ex.map(one, versions))
+            infos = list(ex.map(info, versions))
+        results: Dict[str, Dict[str, "torch.Tensor"]] = {v: {} for v in versions}
+        zstd_jobs = []
+        plain_jobs = []
+        for v, manifest, plan_blobs in infos:
+            for d in manifest.blobs:
+                if not d.size or d.media_type == types.MEDIA_TYPE_MODEL_LEAVES:
+                    continue
+                entry = plan_blobs.get(d.digest)
+                if d.media_type == types.MEDIA_TYPE_MODEL_FILE_ZSTD:
+                    zstd_jobs.append(((v, d.name), d, entry))
+                else:
+                    plain_jobs.append((v, d, entry))
+        if len(zstd_jobs) > 1:
+            for (v, name), t in self._pull_zstd_batched(repository, zstd_jobs,
+                                                        verify, parallel).items():
+                results[v][name] = t
+        elif zstd_jobs:
+            (v, name), d, entry = zstd_jobs[0]
+            results[v][name] = self.pull_zstd_blob_to_device(repository, d,
+                                                             verify=verify,
+                                                             plan_entry=entry)
+
+        def one(job):
+            v, d, entry = job
+            return v, d.name, self.pull_blob_to_device(repository, d, verify=verify,
+                                                       plan_entry=entry)
+
+        if plain_jobs:
+            with ThreadPoolExecutor(max_workers=parallel) as ex:
+                for v, name, t in ex.map(one, plain_jobs):
+                    results[v][name] = t
+        return results
 
     # -------------------------------------------------- directory blobs --
 
@@ -662,16 +807,29 @@ class GpuClient:
         if loc is None:
             # redirect-less registry: direct PUT through the registry
             # (reference pushBlob fallback, push.go:196-207), streamed D2H
-            # in bounded slot-sized pieces
-            def body():
-                off = 0
-                while off < size:
-                    n = min(self.slot_bytes, size - off)
-                    yield self.engine.read_device(ptr + off, n)
-                    off += n
+            # in bounded slot-sized pieces. The reader carries a length so
+            # requests sends Content-Length (the registry, like S3, takes
+            # identity bodies, not chunked transfer encoding).
+            engine, slot = self.engine, self.slot_bytes
+
+            class _DeviceReader:
+                def __init__(self):
+                    self.len = size
+                    self._off = 0
+                    self._buf = b""
+
+                def read(self, n=-1):
+                    if n is None or n < 0:
+                        n = size - self._off + len(self._buf)
+                    while len(self._buf) < n and self._off < size:
+                        take = min(slot, size - self._off)
+                        self._buf += engine.read_device(ptr + self._off, take)
+                        self._off += take
+                    out, self._buf = self._buf[:n], self._buf[n:]
+                    return out
 
             self.remote.upload_blob_content(repository, desc,
-                                            body() if size else b"")
+                                            _DeviceReader() if size else b"")
             self.last_stats.append({"phase": "push-registry-stream", "bytes": size,
                                     "seconds": time.monotonic() - t0})
             return

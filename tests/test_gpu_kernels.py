@@ -442,22 +442,28 @@ class TestPushFaultInjection:
         mdx = start_modelxd_s3(s3d.url, redirect=True)
         try:
             g = GpuClient(mdx.url, device=0, num_slots=4, slot_bytes=8 << 20)
-            real = g.engine.push_part_from_device
             state = {"fails": 0}
+            real_engine = g.engine
 
-            def flaky(url, method, headers, ptr, length):
-                if state["fails"] < 2:
-                    state["fails"] += 1
-                    raise RuntimeError("push: send_body failed (injected)")
-                return real(url, method, headers, ptr, length)
+            class FlakyEngine:
+                """pybind11 instances reject setattr — proxy the engine and
+                fail the first two part uploads at the transport level."""
 
-            monkeypatch.setattr(g.engine, "push_part_from_device", flaky,
-                                raising=False)
+                def __getattr__(self, name):
+                    return getattr(real_engine, name)
+
+                def push_part_from_device(self, url, method, headers, ptr, length):
+                    if state["fails"] < 2:
+                        state["fails"] += 1
+                        raise RuntimeError("push: send_body failed (injected)")
+                    return real_engine.push_part_from_device(url, method, headers,
+                                                             ptr, length)
+
+            g.engine = FlakyEngine()
             src = torch.randint(0, 256, (20 << 20,), dtype=torch.uint8,
                                 device="cuda:0")
             g.push_from_gpu("gpu/fault", "v1", {"w.bin": src}, part_bytes=4 << 20)
-            monkeypatch.setattr(g.engine, "push_part_from_device", real,
-                                raising=False)
+            g.engine = real_engine
             assert state["fails"] == 2
             back = g.pull_to_gpu("gpu/fault", "v1")
             assert torch.equal(back["w.bin"], src)
@@ -487,3 +493,57 @@ class TestRegistryStreamFallback:
         finally:
             mdx.stop()
             s3d.stop()
+
+
+class TestZstdBatchedPull:
+    def test_many_zstd_blobs_batched_decode(self, tmp_path):
+        """pull_to_gpu with several +zstd blobs runs the batched decode
+        (one footer/table pass + one rc sync for the whole set) and the
+        batched raw-digest verify; bytes must round-trip exactly."""
+        from util_servers import start_modelxd_s3, start_s3d
+
+        from modelx_amd.client.gpu import GpuClient
+
+        s3d = start_s3d(str(tmp_path / "s3"))
+        mdx = start_modelxd_s3(s3d.url, redirect=True)
+        try:
+            g = GpuClient(mdx.url, device=0, num_slots=4, slot_bytes=8 << 20)
+            tensors = {}
+            for i in range(6):
+                data = (bytes([i]) * 997 + os.urandom(31)) * 4096  # compressible
+                tensors[f"w{i}.bin"] = torch.frombuffer(bytearray(data),
+                                                        dtype=torch.uint8).cuda()
+            g.push_from_gpu("gpu/zmany", "v1", tensors, compress="zstd")
+            back = g.pull_to_gpu("gpu/zmany", "v1", parallel=4)
+            for name, t in tensors.items():
+                assert torch.equal(back[name], t), name
+            phases = [s["phase"] for s in g.last_stats]
+            assert "pull-zstd-decompress-batched" in phases
+            assert "pull-zstd-raw-verify-batched" in phases
+            # corruption must still surface through the batched path
+            import modelx_amd.wire.errors as er2
+
+            store_root = s3d.proc.args[s3d.proc.args.index("--root") + 1]
+            import glob as g2
+
+            blobs = g2.glob(os.path.join(store_root, "modelx", "registry",
+                                         "gpu/zmany/blobs/*/*"))
+            big = max(blobs, key=os.path.getsize)
+            with open(big, "r+b") as f:
+                f.seek(os.path.getsize(big) // 3)
+                b0 = f.read(1)
+                f.seek(os.path.getsize(big) // 3)
+                f.write(bytes([b0[0] ^ 0x40]))
+            with pytest.raises(Exception):
+                g.pull_to_gpu("gpu/zmany", "v1", parallel=4)
+        finally:
+            mdx.stop()
+            s3d.stop()
+
+    def test_leaves_many_matches_single(self, engine):
+        datas = [torch.randint(0, 256, (s,), dtype=torch.uint8).cuda()
+                 for s in (1, 4096, (1 << 20) + 17, 5 << 20)]
+        items = [(t.data_ptr(), t.numel(), 1 << 20) for t in datas]
+        many = engine.sha256_chunk_leaves_many(items)
+        for t, leaves in zip(datas, many):
+            assert leaves == engine.sha256_chunk_leaves(t.data_ptr(), t.numel(), 1 << 20)
