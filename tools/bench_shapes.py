@@ -49,6 +49,50 @@ def start_stack(rank: int, store_root: str):
     return procs, f"http://127.0.0.1:{mdx_port}"
 
 
+def config3(args, g, dist, rank, world, device):
+    """BASELINE config 3: push 64×1 GiB random blobs via presigned-URL
+    multipart, GPU digests, 1 MI355X. The digest phase is the batched
+    push_from_gpu path: one sha256_chunk_leaves_many call + (canonical
+    mode) one sha256_multibuf call — 64 canonical chains in parallel on
+    device, the stated purpose of the multibuf kernel. Timed region =
+    digest + upload + manifest PUT (the full push)."""
+    import torch
+
+    n_blobs = max(int(64 * args.scale), 2)
+    blob_sz = 1 << 30
+    tensors = {}
+    for i in range(n_blobs):
+        t = torch.randint(0, 256, (blob_sz,), dtype=torch.uint8,
+                          device=f"cuda:{device}")
+        tensors[f"b{i:03d}.bin"] = t
+    torch.cuda.synchronize(device)
+    out = {}
+    for mode in ("chunked", "sha256"):
+        g.last_stats.clear()
+        t0 = time.monotonic()
+        for s in range(args.steps):
+            g.push_from_gpu(f"bench/cfg3-{mode}", f"s{s}", tensors,
+                            digest_mode=mode)
+            g.remote.delete_index(f"bench/cfg3-{mode}")
+        dt = time.monotonic() - t0
+        import collections
+
+        agg = collections.defaultdict(lambda: [0.0, 0])
+        for s in g.last_stats:
+            agg[s.get("phase", "?")][0] += s.get("seconds", 0.0)
+            agg[s.get("phase", "?")][1] += s.get("bytes", 0)
+        out[mode] = {
+            "push_gib_per_s": round(n_blobs * args.steps / dt, 3),
+            "wall_s": round(dt, 2),
+            "stages": {k: {"s": round(v[0], 2), "gib": round(v[1] / (1 << 30), 2),
+                           "gib_per_s": round(v[1] / v[0] / (1 << 30), 1) if v[0] else 0}
+                       for k, v in sorted(agg.items())},
+        }
+    return {"metric": "config3 64x1GiB multipart push GiB/s (GPU digests)",
+            "n_blobs": n_blobs, "value": out["chunked"]["push_gib_per_s"],
+            "modes": out}
+
+
 def config4(args, g, dist, rank, world, device):
     """Sharded Llama-70B-shape: 30 shards, 160 GiB at scale 1."""
     import torch
@@ -198,7 +242,7 @@ def config5(args, g, dist, rank, world, device):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("config", choices=["config4", "config5"])
+    ap.add_argument("config", choices=["config3", "config4", "config5"])
     ap.add_argument("--scale", type=float, default=0.02,
                     help="1.0 = full BASELINE shape (160 GiB / 2 TiB per node)")
     ap.add_argument("--steps", type=int, default=1)
@@ -226,7 +270,7 @@ def main():
 
     g = GpuClient(url, device=device, num_conns=args.conns)
     try:
-        fn = config4 if args.config == "config4" else config5
+        fn = {"config3": config3, "config4": config4, "config5": config5}[args.config]
         result = fn(args, g, dist, rank, world, device)
         result.update({"n_gpus": world, "scale": args.scale, "steps": args.steps})
         if rank == 0:
