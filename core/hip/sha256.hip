@@ -113,14 +113,11 @@ DEV void store_digest(const Sha256State& st, uint8_t* out32) {
 
 constexpr int kBlockThreads = 256;  // 4 wavefronts
 
-// One chunk per lane. chunk c covers [c*chunk_size, min((c+1)*chunk_size,
-// total)); leaves: 32 B per chunk.
-__global__ __launch_bounds__(kBlockThreads) void sha256_chunk_leaves_kernel(
-    const uint8_t* __restrict__ data, uint64_t total, uint64_t chunk_size,
-    uint32_t nchunks, uint8_t* __restrict__ leaves) {
-  __shared__ uint32_t lds_tail[kBlockThreads * kTailDwords];
-  const uint32_t chunk = blockIdx.x * kBlockThreads + threadIdx.x;
-  if (chunk >= nchunks) return;
+// Hash chunk `chunk` of one buffer (one lane) — shared by the single- and
+// many-buffer chunk-leaf kernels.
+DEV void hash_one_chunk(const uint8_t* __restrict__ data, uint64_t total,
+                        uint64_t chunk_size, uint32_t chunk,
+                        uint8_t* __restrict__ leaves, uint32_t* lds_tail_slot) {
   const uint64_t begin = static_cast<uint64_t>(chunk) * chunk_size;
   const uint64_t len = (begin + chunk_size <= total) ? chunk_size : (total - begin);
   const uint8_t* p = data + begin;
@@ -152,9 +149,50 @@ __global__ __launch_bounds__(kBlockThreads) void sha256_chunk_leaves_kernel(
       sha256_compress(st, W);
     }
   }
-  sha256_tail(st, p + (nfull << 6), len - (nfull << 6), len,
-              &lds_tail[threadIdx.x * kTailDwords]);
+  sha256_tail(st, p + (nfull << 6), len - (nfull << 6), len, lds_tail_slot);
   store_digest(st, leaves + static_cast<uint64_t>(chunk) * 32);
+}
+
+// One chunk per lane. chunk c covers [c*chunk_size, min((c+1)*chunk_size,
+// total)); leaves: 32 B per chunk.
+__global__ __launch_bounds__(kBlockThreads) void sha256_chunk_leaves_kernel(
+    const uint8_t* __restrict__ data, uint64_t total, uint64_t chunk_size,
+    uint32_t nchunks, uint8_t* __restrict__ leaves) {
+  __shared__ uint32_t lds_tail[kBlockThreads * kTailDwords];
+  const uint32_t chunk = blockIdx.x * kBlockThreads + threadIdx.x;
+  if (chunk >= nchunks) return;
+  hash_one_chunk(data, total, chunk_size, chunk, leaves,
+                 &lds_tail[threadIdx.x * kTailDwords]);
+}
+
+// Chunk leaves of MANY buffers in ONE launch: a 64 MiB blob alone is 512
+// chains (a near-idle chip, ~2 ms per launch × N blobs serialized on one
+// stream — the measured config-5 verify wall); the combined grid puts every
+// blob's chunks in flight at once. Buffers are located by binary search
+// over the chunk-prefix array in the descriptors.
+struct ChunkLeavesDesc {
+  const uint8_t* data;
+  uint8_t* leaves;
+  uint64_t total;
+  uint64_t chunk_size;
+  uint32_t chunk_base;  // chunks in all preceding buffers
+  uint32_t pad_;
+};
+
+__global__ __launch_bounds__(kBlockThreads) void sha256_chunk_leaves_many_kernel(
+    const ChunkLeavesDesc* __restrict__ descs, uint32_t nbuf, uint32_t total_chunks) {
+  __shared__ uint32_t lds_tail[kBlockThreads * kTailDwords];
+  const uint32_t g = blockIdx.x * kBlockThreads + threadIdx.x;
+  if (g >= total_chunks) return;
+  uint32_t lo = 0, hi = nbuf - 1;
+  while (lo < hi) {
+    uint32_t mid = (lo + hi + 1) >> 1;
+    if (descs[mid].chunk_base <= g) lo = mid;
+    else hi = mid - 1;
+  }
+  const ChunkLeavesDesc d = descs[lo];
+  hash_one_chunk(d.data, d.total, d.chunk_size, g - d.chunk_base, d.leaves,
+                 &lds_tail[threadIdx.x * kTailDwords]);
 }
 
 // Canonical single-stream SHA-256 of N independent buffers, one per lane:
@@ -214,6 +252,15 @@ hipError_t modelx_sha256_chunk_leaves(const void* data, uint64_t total, uint64_t
   hipLaunchKernelGGL(sha256_chunk_leaves_kernel, grid, dim3(kBlockThreads), 0, stream,
                      static_cast<const uint8_t*>(data), total, chunk_size, nchunks,
                      static_cast<uint8_t*>(leaves));
+  return hipGetLastError();
+}
+
+hipError_t modelx_sha256_chunk_leaves_many(const void* descs_dev, uint32_t nbuf,
+                                           uint32_t total_chunks, hipStream_t stream) {
+  if (total_chunks == 0 || nbuf == 0) return hipSuccess;
+  dim3 grid((total_chunks + kBlockThreads - 1) / kBlockThreads);
+  hipLaunchKernelGGL(sha256_chunk_leaves_many_kernel, grid, dim3(kBlockThreads), 0, stream,
+                     static_cast<const ChunkLeavesDesc*>(descs_dev), nbuf, total_chunks);
   return hipGetLastError();
 }
 

@@ -51,6 +51,19 @@ extern "C" hipError_t modelx_sha256_chunk_leaves(const void* data, uint64_t tota
                                                  uint32_t nchunks, hipStream_t stream);
 extern "C" hipError_t modelx_sha256_multibuf(const void* const* buffers, const uint64_t* lengths,
                                              uint32_t nbuf, void* digests, hipStream_t stream);
+extern "C" hipError_t modelx_sha256_chunk_leaves_many(const void* descs_dev, uint32_t nbuf,
+                                                      uint32_t total_chunks,
+                                                      hipStream_t stream);
+
+// host mirror of sha256.hip ChunkLeavesDesc
+struct ChunkLeavesDescHost {
+  const void* data;
+  void* leaves;
+  uint64_t total;
+  uint64_t chunk_size;
+  uint32_t chunk_base;
+  uint32_t pad_;
+};
 extern "C" hipError_t modelx_tar_index(const void* tar, uint64_t tar_len, void* entries,
                                        uint32_t max_entries, uint32_t* count_dev,
                                        uint32_t* error_dev, hipStream_t stream);
@@ -1076,12 +1089,20 @@ class GpuEngine {
       py::gil_scoped_release release;
       std::lock_guard<std::mutex> zlk(zstd_mu_);
       void* dleaves = zstd_scratch(8, total * 32);
+      // ONE launch over all buffers: per-blob launches leave the chip
+      // near-idle on small blobs (a 64 MiB blob is only 512 chains)
+      std::vector<ChunkLeavesDescHost> descs(n);
       for (size_t i = 0; i < n; i++) {
         auto [ptr, size, cs] = items[i];
-        HIP_CHECK(modelx_sha256_chunk_leaves(reinterpret_cast<void*>(ptr), size, cs,
-                                             static_cast<char*>(dleaves) + off[i] * 32,
-                                             nchunks[i], hash_stream_));
+        descs[i] = {reinterpret_cast<const void*>(ptr),
+                    static_cast<char*>(dleaves) + off[i] * 32, size, cs,
+                    (uint32_t)off[i], 0};
       }
+      void* ddescs = zstd_scratch(9, n * sizeof(ChunkLeavesDescHost));
+      HIP_CHECK(hipMemcpyAsync(ddescs, descs.data(), n * sizeof(ChunkLeavesDescHost),
+                               hipMemcpyHostToDevice, hash_stream_));
+      HIP_CHECK(modelx_sha256_chunk_leaves_many(ddescs, (uint32_t)n, (uint32_t)total,
+                                                hash_stream_));
       HIP_CHECK(hipStreamSynchronize(hash_stream_));
       host.resize(total * 32);
       HIP_CHECK(hipMemcpy(&host[0], dleaves, host.size(), hipMemcpyDeviceToHost));

@@ -210,15 +210,16 @@ def test_fanout_broadcast_rccl_world2():
     import sys
 
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1", NCCL_DEBUG="WARN")
     env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29881",
+           "--master-port", "29881", "--log-dir", "/tmp/rccl2-logs",
+           "--redirects", "3", "--tee", "3",
            os.path.join(repo, "tests", "rccl2_helper.py")]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
                        cwd=repo, env=env)
-    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "RCCL2 OK" in r.stdout
 
 

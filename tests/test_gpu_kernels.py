@@ -416,6 +416,10 @@ class TestCanonicalDigest:
 
         size = 256 << 20
         dev = torch.randint(0, 256, (size,), dtype=torch.uint8, device="cuda")
+        # raw engine calls never order against torch's stream — sync the
+        # producer like GpuClient._sync_producers does, or the D2H reads
+        # race the randint kernel (observed as a digest mismatch on HW)
+        torch.cuda.synchronize()
         t0 = time.monotonic()
         got = engine.sha256_canonical_device(dev.data_ptr(), size)
         t_cpu = time.monotonic() - t0
