@@ -6,7 +6,9 @@
 #include <signal.h>
 #include <unistd.h>
 
+#include <atomic>
 #include <chrono>
+#include <thread>
 #include <cstdio>
 #include <cstring>
 #include <memory>
@@ -25,6 +27,7 @@ int main(int argc, char** argv) {
   bool enable_redirect = false;
   registry::AuthConfig auth;
   http::TlsConfig tls;
+  int gc_interval_s = 0;
 
   for (int i = 1; i < argc; i++) {
     std::string a = argv[i];
@@ -58,6 +61,7 @@ int main(int argc, char** argv) {
       }
     } else if (a == "--oidc-issuer") auth.oidc_issuer = next();
     else if (a == "--oidc-audience") auth.oidc_audience = next();
+    else if (a == "--gc-interval") gc_interval_s = atoi(next().c_str());
     else if (a == "--tls-cert") tls.cert_file = next();
     else if (a == "--tls-key") tls.key_file = next();
     else if (a == "--help" || a == "-h") {
@@ -74,6 +78,8 @@ int main(int argc, char** argv) {
              "  --oidc-jwks FILE          verify RS256 ID tokens against a JWKS document\n"
              "  --oidc-issuer ISS         require `iss` claim to equal ISS\n"
              "  --oidc-audience AUD       require `aud` claim to contain AUD\n"
+             "  --gc-interval SECS        periodic mark-sweep GC of all repositories\n"
+             "                            (reference has manual POST garbage-collect only)\n"
              "  --tls-cert F --tls-key F  serve HTTPS (reference --tls-*, server.go:37-43)\n");
       return 0;
     }
@@ -111,6 +117,22 @@ int main(int argc, char** argv) {
   printf("modelxd listening on port %d backend=%s redirect=%d\n", port,
          use_s3 ? "s3" : "local", enable_redirect ? 1 : 0);
   fflush(stdout);
+  // scheduled GC sweep (reference ships manual POST garbage-collect only;
+  // operational gap noted in docs/roadmap.md). Runs the same mark-sweep the
+  // endpoint runs; content-addressed writes make sweeps safe to repeat.
+  std::atomic<bool> gc_stop{false};
+  std::thread gc_thread;
+  if (gc_interval_s > 0) {
+    gc_thread = std::thread([&] {
+      while (!gc_stop.load()) {
+        for (int i = 0; i < gc_interval_s * 10 && !gc_stop.load(); i++)
+          std::this_thread::sleep_for(std::chrono::milliseconds(100));
+        if (gc_stop.load()) break;
+        int removed = st->GCBlobsAll();
+        fprintf(stderr, "gc sweep: %d blob(s) removed\n", removed);
+      }
+    });
+  }
   sigset_t set;
   sigemptyset(&set);
   sigaddset(&set, SIGINT);
@@ -118,6 +140,8 @@ int main(int argc, char** argv) {
   sigprocmask(SIG_BLOCK, &set, nullptr);
   int sig = 0;
   sigwait(&set, &sig);
+  gc_stop.store(true);
+  if (gc_thread.joinable()) gc_thread.join();
   server.stop();  // graceful shutdown (server.go:33-36)
   return 0;
 }

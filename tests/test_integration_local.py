@@ -167,3 +167,30 @@ def test_chunked_digest_mode(server, model_dir, tmp_path):
     out = tmp_path / "chunked-out"
     c.pull("proj/chunked", "v1", str(out), quiet=True)
     assert (out / "weights.bin").read_bytes() == (model_dir / "weights.bin").read_bytes()
+
+
+def test_scheduled_gc_sweeps_orphans(model_dir, tmp_path):
+    """--gc-interval runs the mark-sweep periodically (the reference has
+    only the manual POST endpoint)."""
+    import time
+
+    from util_servers import MODELXD, ServerProc, _build_servers, free_port, wait_http
+
+    _build_servers()
+    port = free_port()
+    p = ServerProc([MODELXD, "--listen", f"127.0.0.1:{port}", "--local-data",
+                    str(tmp_path / "reg"), "--gc-interval", "1"], port)
+    wait_http(port)
+    try:
+        c = Client(p.url)
+        c.push("proj/gcsched", "v1", str(model_dir), quiet=True)
+        m = c.get_manifest("proj/gcsched", "v1")
+        orphan = next(b.digest for b in m.blobs if b.name == "weights.bin")
+        m.blobs = [b for b in m.blobs if b.name != "weights.bin"]
+        c.remote.put_manifest("proj/gcsched", "v1", m)
+        deadline = time.time() + 10
+        while time.time() < deadline and c.remote.head_blob("proj/gcsched", orphan):
+            time.sleep(0.3)
+        assert not c.remote.head_blob("proj/gcsched", orphan)
+    finally:
+        p.stop()
