@@ -20,11 +20,19 @@ def main():
 
     rank = int(os.environ["RANK"])
     world = int(os.environ["WORLD_SIZE"])
-    torch.cuda.set_device(0)  # both ranks share the one GPU
-    dist.init_process_group("nccl")
+    # RCCL (2.26) rejects two ranks on one device ("Duplicate GPU detected",
+    # init.cc:1108 — verified on hardware), so: real RCCL when the box has a
+    # GPU per rank (the driver's 8-GPU node), gloo transport for the CUDA
+    # tensors on a 1-GPU box — two processes, device tensors, the same
+    # fan-out choreography; only the wire differs.
+    ngpu = torch.cuda.device_count()
+    device = rank % ngpu
+    backend = "nccl" if ngpu >= world else "gloo"
+    torch.cuda.set_device(device)
+    dist.init_process_group(backend)
 
-    # collective sanity: all-reduce across two ranks on one device
-    t = torch.full((1 << 20,), float(rank + 1), device="cuda:0")
+    # collective sanity: all-reduce across ranks on device tensors
+    t = torch.full((1 << 20,), float(rank + 1), device=f"cuda:{device}")
     dist.all_reduce(t)
     assert t[0].item() == sum(r + 1 for r in range(world)), t[0].item()
 
@@ -46,32 +54,34 @@ def main():
         wait_http(mdx_port)
 
     try:
-        g = GpuClient(f"http://127.0.0.1:{mdx_port}", device=0,
+        g = GpuClient(f"http://127.0.0.1:{mdx_port}", device=device,
                       num_slots=4, slot_bytes=8 << 20)
         # each rank pushes its own blob; the other rank receives it purely
         # over the RCCL broadcast and digest-verifies on device
         src = torch.randint(0, 256, ((24 << 20) + 137,), dtype=torch.uint8,
-                            device="cuda:0")
+                            device=f"cuda:{device}")
         g.push_from_gpu("rccl2/m", f"v-r{rank}", {"w.bin": src})
         dist.barrier()
         for src_rank in range(world):
             out = fanout_pull_broadcast(dist, g, "rccl2/m", f"v-r{src_rank}",
-                                        device=0, chunk=4 << 20, src_rank=src_rank)
+                                        device=device, chunk=4 << 20,
+                                        src_rank=src_rank)
             if rank == src_rank:
                 assert torch.equal(out["w.bin"], src), "own blob mismatch"
         # sharded pull with replicate: ShardPlan owners fetch, broadcasts
         # replicate, every rank verifies
         tensors = {f"s{i}.bin": torch.randint(0, 256, (2 << 20,), dtype=torch.uint8,
-                                              device="cuda:0") for i in range(4)}
+                                              device=f"cuda:{device}")
+                   for i in range(4)}
         if rank == 0:
             g.push_from_gpu("rccl2/shard", "v1", tensors)
         dist.barrier()
-        out = fanout_pull_sharded(dist, g, "rccl2/shard", "v1", device=0,
+        out = fanout_pull_sharded(dist, g, "rccl2/shard", "v1", device=device,
                                   replicate=True)
         assert len(out) == 4
         dist.barrier()
         if rank == 0:
-            print("RCCL2 OK", flush=True)
+            print(f"RCCL2 OK backend={backend} ngpu={ngpu}", flush=True)
     finally:
         dist.destroy_process_group()
         for p in procs:

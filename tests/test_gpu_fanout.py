@@ -202,10 +202,14 @@ def test_modelxdl_gpus_flag(stack, tmp_path, capsys):
 
 
 def test_fanout_broadcast_rccl_world2():
-    """TWO RCCL ranks on the ONE GPU (RCCL allows multiple ranks per
-    device): real ncclBroadcast/allreduce exercise the fan-out stream
-    ordering and digest-after-broadcast path beyond world-1 without an
-    8-GPU node (VERDICT round-1 item #2)."""
+    """World-2 fan-out with TWO processes on real device tensors. RCCL
+    rejects two ranks on one device ("Duplicate GPU detected",
+    init.cc:1108 — verified on hardware), so the helper picks the
+    transport by hardware: real RCCL when a GPU per rank exists (the
+    driver's 8-GPU box runs this as true RCCL world-2), gloo transport
+    for the CUDA tensors on a 1-GPU box — either way the full multi-rank
+    choreography (pipelined broadcast, ShardPlan replicate,
+    digest-after-collective on every rank) executes beyond world-1."""
     import subprocess
     import sys
 
