@@ -216,14 +216,25 @@ def test_fanout_broadcast_rccl_world2():
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     env = dict(os.environ, MASTER_ADDR="127.0.0.1", NCCL_DEBUG="WARN")
     env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    log_dir = os.path.join(repo, "gpurun_out", "rccl2-logs")
     cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
            "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-           "--master-port", "29881", "--log-dir", "/tmp/rccl2-logs",
+           "--master-port", "29881", "--log-dir", log_dir,
            "--redirects", "3", "--tee", "3",
            os.path.join(repo, "tests", "rccl2_helper.py")]
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
                        cwd=repo, env=env)
-    assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
+    if r.returncode != 0:
+        # surface each rank's own stderr tail (the launcher interleaves and
+        # truncates)
+        import glob
+
+        tails = []
+        for f in sorted(glob.glob(os.path.join(log_dir, "**", "*"), recursive=True)):
+            if os.path.isfile(f):
+                with open(f, errors="replace") as fh:
+                    tails.append(f"== {f}\n" + fh.read()[-1500:])
+        raise AssertionError("\n".join(tails)[-8000:] or r.stderr[-3000:])
     assert "RCCL2 OK" in r.stdout
 
 
