@@ -551,3 +551,36 @@ class TestZstdBatchedPull:
         many = engine.sha256_chunk_leaves_many(items)
         for t, leaves in zip(datas, many):
             assert leaves == engine.sha256_chunk_leaves(t.data_ptr(), t.numel(), 1 << 20)
+
+
+class TestFseWeightsDevice:
+    def test_device_compress_bf16_bytes(self, engine):
+        """The CDNA4 compress kernel emits FSE-compressed huffman weight
+        tables for full-byte alphabets (bf16 tensor bytes): ratio must beat
+        raw and the blob must decode via our decoder AND libzstd."""
+        import ctypes
+
+        from modelx_amd import _core
+
+        torch.manual_seed(5)
+        data = ((torch.randn(4 << 20) * 0.02).to(torch.bfloat16)
+                .view(torch.uint8).numpy().tobytes())
+        src = torch.frombuffer(bytearray(data), dtype=torch.uint8).cuda()
+        bound = _core.zstd_compress_bound(len(data))
+        dst = torch.empty(bound, dtype=torch.uint8, device="cuda")
+        n = engine.zstd_compress_device(src.data_ptr(), len(data), 128 << 10,
+                                        dst.data_ptr(), bound)
+        blob = bytes(dst[:n].cpu().numpy().tobytes())
+        assert n / len(data) < 0.85, f"ratio {n / len(data):.3f}"
+        assert _core.zstd_decompress_cpu(blob) == data
+        back = torch.empty(len(data), dtype=torch.uint8, device="cuda")
+        m = engine.zstd_decompress_device(dst.data_ptr(), n, back.data_ptr(),
+                                          len(data))
+        assert m == len(data)
+        assert bytes(back.cpu().numpy().tobytes()) == data
+        z = ctypes.CDLL("libzstd.so.1")
+        z.ZSTD_decompress.restype = ctypes.c_size_t
+        z.ZSTD_isError.restype = ctypes.c_uint
+        out = ctypes.create_string_buffer(len(data))
+        k = z.ZSTD_decompress(out, len(data), blob, len(blob))
+        assert not z.ZSTD_isError(k) and out.raw[:k] == data
