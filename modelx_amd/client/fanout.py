@@ -150,7 +150,8 @@ def fanout_pull_sharded(dist, gpu_client, repository: str, version: str, device:
     rank = dist.get_rank()
     world = dist.get_world_size()
     manifest = gpu_client.remote.get_manifest(repository, version)
-    descs = [d for d in manifest.blobs if d.size > 0]
+    descs = [d for d in manifest.blobs
+             if d.size > 0 and d.media_type != types.MEDIA_TYPE_MODEL_LEAVES]
     plan = ShardPlan.build(descs, world)
     out = {}
     owned = [d for d in descs if plan.owners[d.name] == rank]
