@@ -127,6 +127,10 @@ def fanout_pull_broadcast(dist, gpu_client, repository: str, version: str,
         broadcast_blob_pipelined(dist, tensor, desc.size, src_rank, fetch, chunk)
         if verify:
             if is_gpu:
+                # the collective is async on torch's stream; the verify
+                # kernel runs on the engine's own stream with no implicit
+                # ordering — sync or the digest reads torn bytes
+                torch.cuda.synchronize(device)
                 gpu_client._verify_device_digest(tensor.data_ptr(), desc.size, desc)
             else:
                 from ..wire import digest as dg
@@ -178,6 +182,7 @@ def fanout_pull_sharded(dist, gpu_client, repository: str, version: str, device:
             tensor = out[desc.name]
             broadcast_blob_pipelined(dist, tensor, desc.size, plan.owners[desc.name], None)
             if verify and plan.owners[desc.name] != rank:
+                torch.cuda.synchronize(device)  # order collective before engine read
                 gpu_client._verify_device_digest(tensor.data_ptr(), desc.size, desc)
     return out
 
