@@ -25,3 +25,11 @@ for _ in range(reps):
     eng.sha256_chunk_leaves(buf.data_ptr(), size, cs)
 dt = (time.monotonic() - t0) / reps
 print(f"sha256_chunk_leaves: {size / dt / (1 << 30):.1f} GiB/s (chunk={cs}, reps={reps})")
+
+# small-blob regime A/B: 64 MiB spans (the per-slot streaming-hash shape)
+small = 64 << 20
+t0 = time.monotonic()
+for _ in range(20):
+    eng.sha256_chunk_leaves(buf.data_ptr(), small, cs)
+dt = (time.monotonic() - t0) / 20
+print(f"sha256_chunk_leaves 64MiB: {small / dt / (1 << 30):.1f} GiB/s")
