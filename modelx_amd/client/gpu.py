@@ -516,7 +516,7 @@ class GpuClient:
 
         import torch
 
-        wave_cap = 64 << 30
+        wave_cap = self.ZSTD_WAVE_CAP
         if len(jobs) > 1:
             total_raw = sum(int(d.annotations.get(types.ANNOTATION_RAW_SIZE, 0) or 0)
                             for _, d, _ in jobs)
@@ -826,6 +826,7 @@ class GpuClient:
         return root, leaves
 
     PART_RETRIES = 3  # reference: pkg/client/extension_s3.go:133-148
+    ZSTD_WAVE_CAP = 64 << 30  # raw bytes per batched-decode wave
 
     def _upload_small(self, repository: str, desc: types.Descriptor, data: bytes) -> None:
         loc = self.remote.get_blob_location(repository, desc, "upload")
