@@ -113,101 +113,6 @@ DEV void store_digest(const Sha256State& st, uint8_t* out32) {
 
 constexpr int kBlockThreads = 256;  // 4 wavefronts
 
-// 64 rounds over kIlp INDEPENDENT blocks, rounds interleaved across the
-// chains so consecutive VALU instructions have no dependency. At small
-// grids (a 64 MiB blob = 512 chains = 8 waves on a 1024-SIMD chip) there
-// is no wave co-residency to hide the ~4-cycle round dependency, and the
-// plain kernel runs at ~0.25 IPC (measured 9 GiB/s per 64 MiB launch);
-// in-lane ILP restores ~1 IPC without needing more chains.
-constexpr int kIlp = 4;
-
-DEV void sha256_compress_ilp(Sha256State* st, uint32_t W[kIlp][16]) {
-  uint32_t a[kIlp], b[kIlp], c[kIlp], d[kIlp], e[kIlp], f[kIlp], g[kIlp], hh[kIlp];
-#pragma unroll
-  for (int j = 0; j < kIlp; j++) {
-    a[j] = st[j].h[0]; b[j] = st[j].h[1]; c[j] = st[j].h[2]; d[j] = st[j].h[3];
-    e[j] = st[j].h[4]; f[j] = st[j].h[5]; g[j] = st[j].h[6]; hh[j] = st[j].h[7];
-  }
-#pragma unroll
-  for (int t = 0; t < 64; t++) {
-    uint32_t k = K256[t];
-#pragma unroll
-    for (int j = 0; j < kIlp; j++) {
-      uint32_t w;
-      if (t < 16) {
-        w = W[j][t];
-      } else {
-        uint32_t w15 = W[j][(t - 15) & 15], w2 = W[j][(t - 2) & 15];
-        uint32_t s0 = rotr(w15, 7) ^ rotr(w15, 18) ^ (w15 >> 3);
-        uint32_t s1 = rotr(w2, 17) ^ rotr(w2, 19) ^ (w2 >> 10);
-        w = W[j][t & 15] + s0 + W[j][(t - 7) & 15] + s1;
-        W[j][t & 15] = w;
-      }
-      uint32_t S1 = rotr(e[j], 6) ^ rotr(e[j], 11) ^ rotr(e[j], 25);
-      uint32_t ch = (e[j] & f[j]) ^ (~e[j] & g[j]);
-      uint32_t t1 = hh[j] + S1 + ch + k + w;
-      uint32_t S0 = rotr(a[j], 2) ^ rotr(a[j], 13) ^ rotr(a[j], 22);
-      uint32_t maj = (a[j] & b[j]) ^ (a[j] & c[j]) ^ (b[j] & c[j]);
-      uint32_t t2 = S0 + maj;
-      hh[j] = g[j]; g[j] = f[j]; f[j] = e[j]; e[j] = d[j] + t1;
-      d[j] = c[j]; c[j] = b[j]; b[j] = a[j]; a[j] = t1 + t2;
-    }
-  }
-#pragma unroll
-  for (int j = 0; j < kIlp; j++) {
-    st[j].h[0] += a[j]; st[j].h[1] += b[j]; st[j].h[2] += c[j]; st[j].h[3] += d[j];
-    st[j].h[4] += e[j]; st[j].h[5] += f[j]; st[j].h[6] += g[j]; st[j].h[7] += hh[j];
-  }
-}
-
-// kIlp FULL chunks per lane (the host peels short tails to the plain
-// kernel). 64-thread blocks: the regime is few-wave by definition.
-__global__ __launch_bounds__(64) void sha256_chunk_leaves_ilp_kernel(
-    const uint8_t* __restrict__ data, uint64_t chunk_size, uint32_t base_chunk,
-    uint32_t ngroups, uint8_t* __restrict__ leaves) {
-  uint32_t gidx = blockIdx.x * 64 + threadIdx.x;
-  if (gidx >= ngroups) return;
-  uint64_t c0 = (uint64_t)base_chunk + (uint64_t)gidx * kIlp;
-  Sha256State st[kIlp];
-  const uint8_t* p[kIlp];
-#pragma unroll
-  for (int j = 0; j < kIlp; j++) {
-    st[j].init();
-    p[j] = data + (c0 + j) * chunk_size;
-  }
-  const uint64_t nfull = chunk_size >> 6;  // chunk_size % 64 == 0 (host-gated)
-  uint32_t W[kIlp][16];
-  for (uint64_t blk = 0; blk < nfull; blk++) {
-#pragma unroll
-    for (int j = 0; j < kIlp; j++) {
-      const uint4* v = reinterpret_cast<const uint4*>(p[j]) + blk * 4;  // 16-aligned (host-gated)
-#pragma unroll
-      for (int i = 0; i < 4; i++) {
-        uint4 q = v[i];
-        W[j][i * 4 + 0] = bswap(q.x);
-        W[j][i * 4 + 1] = bswap(q.y);
-        W[j][i * 4 + 2] = bswap(q.z);
-        W[j][i * 4 + 3] = bswap(q.w);
-      }
-    }
-    sha256_compress_ilp(st, W);
-  }
-  // padding block — identical for every full chunk: 0x80, zeros, 64-bit
-  // big-endian bit length (fits one block since chunk_size % 64 == 0)
-  const uint64_t bits = chunk_size << 3;
-#pragma unroll
-  for (int j = 0; j < kIlp; j++) {
-#pragma unroll
-    for (int i = 0; i < 16; i++) W[j][i] = 0;
-    W[j][0] = 0x80000000u;
-    W[j][14] = (uint32_t)(bits >> 32);
-    W[j][15] = (uint32_t)bits;
-  }
-  sha256_compress_ilp(st, W);
-#pragma unroll
-  for (int j = 0; j < kIlp; j++) store_digest(st[j], leaves + (c0 + j) * 32);
-}
-
 // Hash chunk `chunk` of one buffer (one lane) — shared by the single- and
 // many-buffer chunk-leaf kernels.
 DEV void hash_one_chunk(const uint8_t* __restrict__ data, uint64_t total,
@@ -343,34 +248,16 @@ extern "C" {
 hipError_t modelx_sha256_chunk_leaves(const void* data, uint64_t total, uint64_t chunk_size,
                                       void* leaves, uint32_t nchunks, hipStream_t stream) {
   if (nchunks == 0) return hipSuccess;
-  // small-grid regime (< ~1024 waves): not enough chains for wave
-  // co-residency to hide the round dependency — switch to the in-lane
-  // ILP kernel for the full chunks and peel the tail to the plain one
-  bool ilp_ok = chunk_size >= 64 && (chunk_size & 63) == 0 &&
-                (reinterpret_cast<uintptr_t>(data) & 15) == 0 && nchunks >= 2 * kIlp &&
-                nchunks <= (1u << 17);
-  uint32_t done = 0;
-  if (ilp_ok) {
-    uint32_t full = (total % chunk_size == 0) ? nchunks : nchunks - 1;
-    uint32_t groups = full / kIlp;
-    if (groups) {
-      dim3 grid((groups + 63) / 64);
-      hipLaunchKernelGGL(sha256_chunk_leaves_ilp_kernel, grid, dim3(64), 0, stream,
-                         static_cast<const uint8_t*>(data), chunk_size, 0u, groups,
-                         static_cast<uint8_t*>(leaves));
-      hipError_t e = hipGetLastError();
-      if (e != hipSuccess) return e;
-      done = groups * kIlp;
-    }
-  }
-  if (done < nchunks) {
-    uint32_t rem = nchunks - done;
-    dim3 grid((rem + kBlockThreads - 1) / kBlockThreads);
-    hipLaunchKernelGGL(sha256_chunk_leaves_kernel, grid, dim3(kBlockThreads), 0, stream,
-                       static_cast<const uint8_t*>(data) + (uint64_t)done * chunk_size,
-                       total - (uint64_t)done * chunk_size, chunk_size, rem,
-                       static_cast<uint8_t*>(leaves) + (uint64_t)done * 32);
-  }
+  // NOTE: an in-lane ILP4 variant (4 interleaved chains per lane, meant to
+  // hide the round dependency at few-wave grids) was A/B-tested on
+  // hardware and LOST badly: 160 GiB/s on 8 GiB (vs 1017 plain) and
+  // 1.3 GiB/s on 64 MiB (vs 9) — W[4][16] + 4 states under the unrolled
+  // round loop spills to scratch and every round pays a memory round
+  // trip. One chain per lane stays (docs/engineering-notes.md #13).
+  dim3 grid((nchunks + kBlockThreads - 1) / kBlockThreads);
+  hipLaunchKernelGGL(sha256_chunk_leaves_kernel, grid, dim3(kBlockThreads), 0, stream,
+                     static_cast<const uint8_t*>(data), total, chunk_size, nchunks,
+                     static_cast<uint8_t*>(leaves));
   return hipGetLastError();
 }
 
