@@ -963,10 +963,9 @@ class GpuClient:
             self._upload_small(repository, manifest.config, cfg)
         if compress == "" and len(tensors) > 1:
             # many-blob fast path (BASELINE config 3 shape): ONE batched
-            # leaf-digest call for all tensors, canonical digests via the
-            # multibuf kernel (one chain per blob, all blobs in parallel on
-            # GPU — a single canonical chain is CPU-bound, but 64 of them
-            # saturate the device), then blob uploads in parallel threads
+            # leaf-digest call for all tensors, then blob uploads in
+            # parallel threads; canonical mode digests via parallel CPU
+            # SHA-NI chains (see below)
             import time
             from concurrent.futures import ThreadPoolExecutor
 
