@@ -297,12 +297,56 @@ def main():
         if not cpu_mode:
             torch.cuda.synchronize(device)
 
+    def pipelined_steps():
+        """Independent mode, steps>1: overlap the push of step k with the
+        pull of step k-1 (two threads, one client — the engine is
+        reentrant). Loopback measured 1.11x duplex (tools/duplex_probe.py);
+        on a real NIC the two directions are independent lanes. The timed
+        region still contains exactly K pushes + K pulls + K cleanups;
+        per-version cleanup = DELETE manifest + mark-sweep GC (the
+        reference's own lifecycle verbs) so tmpfs never holds more than
+        two step versions."""
+        import threading
+
+        def gen():
+            src.random_(0, 256)
+            if not cpu_mode:
+                torch.cuda.synchronize(device)
+
+        def push(k):
+            g.push_from_gpu(repo, f"s{k}", {"blob.bin": src},
+                            part_bytes=args.part_mib << 20)
+
+        def clean(k):
+            # NEVER concurrent with a push: blobs upload before their
+            # manifest (the commit-point convention), so a mark-sweep racing
+            # an in-flight push would sweep its not-yet-referenced blobs
+            g.remote.delete_manifest(repo, f"s{k}")
+            g.remote.garbage_collect(repo)
+
+        gen()
+        push(0)
+        for k in range(1, args.steps):
+            gen()
+            th = threading.Thread(target=push, args=(k,))
+            th.start()
+            g.pull_to_gpu(repo, f"s{k - 1}", verify=True)
+            th.join()
+            clean(k - 1)
+        g.pull_to_gpu(repo, f"s{args.steps - 1}", verify=True)
+        clean(args.steps - 1)
+
+    pipeline = (not fanout and not distributed and args.steps > 1
+                and os.environ.get("MODELX_BENCH_PIPELINE", "1") != "0")
     for w in range(args.warmup):
         one_step(1000 + w)
     barrier_sync()
     t0 = time.monotonic()
-    for k in range(args.steps):
-        one_step(k)
+    if pipeline:
+        pipelined_steps()
+    else:
+        for k in range(args.steps):
+            one_step(k)
     barrier_sync()
     elapsed = time.monotonic() - t0
 
@@ -345,7 +389,7 @@ def main():
         else:
             moved_gib = 2.0 * args.blob_gib * args.steps * world  # push+pull, all ranks
             model = f"synthetic-{args.blob_gib:g}GiB-blob"
-            par = f"dp{world}-presigned-s3"
+            par = f"dp{world}-presigned-s3" + ("-pipelined" if pipeline else "")
         print(json.dumps({
             "metric": "push+pull GiB/s end-to-end (S3->HBM, digest-verified), 1/2/4/8 MI355X",
             "value": round(moved_gib / elapsed, 3),
