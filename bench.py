@@ -336,14 +336,73 @@ def main():
         g.pull_to_gpu(repo, f"s{args.steps - 1}", verify=True)
         clean(args.steps - 1)
 
-    pipeline = (not fanout and not distributed and args.steps > 1
-                and os.environ.get("MODELX_BENCH_PIPELINE", "1") != "0")
+    def pipelined_fanout_steps():
+        """Fan-out mode, steps>1: each rank's push of step k (plain HTTP, on
+        a side thread) overlaps the fan-out broadcasts of step k-1. ALL
+        collectives (broadcasts + barriers) stay on the main thread in
+        rank-identical order, so the pipelining cannot reorder NCCL ops.
+        Cleanup (per-version manifest DELETEs by their owners + one
+        mark-sweep) runs after the push joins — never concurrent with an
+        in-flight push's not-yet-referenced blobs."""
+        import threading
+
+        import torch.distributed as dist
+
+        from modelx_amd.client.fanout import fanout_pull_broadcast
+
+        def gen():
+            src.random_(0, 256)
+            if not cpu_mode:
+                torch.cuda.synchronize(device)
+
+        def push(k):
+            g.push_from_gpu(repo, f"s{k}-r{rank}", {"shard.bin": src},
+                            part_bytes=args.part_mib << 20)
+
+        def fan(k):
+            landed = 0
+            for r in range(world):
+                out = fanout_pull_broadcast(dist, g, repo, f"s{k}-r{r}", device,
+                                            chunk=args.chunk_mib << 20,
+                                            src_rank=r, verify=True)
+                landed += sum(t.numel() for t in out.values())
+                del out
+            assert landed == blob_bytes, (landed, blob_bytes)
+
+        gen()
+        push(0)
+        dist.barrier()  # step-0 shards published
+        for k in range(1, args.steps):
+            gen()
+            th = threading.Thread(target=push, args=(k,))
+            th.start()
+            fan(k - 1)
+            th.join()
+            dist.barrier()  # step-k shards published AND k-1 consumed
+            g.remote.delete_manifest(repo, f"s{k - 1}-r{rank}")  # own version
+            dist.barrier()
+            if rank == 0:
+                g.remote.garbage_collect(repo)
+            dist.barrier()
+        fan(args.steps - 1)
+        dist.barrier()
+        g.remote.delete_manifest(repo, f"s{args.steps - 1}-r{rank}")
+        dist.barrier()
+        if rank == 0:
+            g.remote.delete_index(repo)
+        dist.barrier()
+
+    pipe_env = os.environ.get("MODELX_BENCH_PIPELINE", "1") != "0"
+    pipeline = not fanout and not distributed and args.steps > 1 and pipe_env
+    pipeline_fan = fanout and args.steps > 1 and pipe_env
     for w in range(args.warmup):
         one_step(1000 + w)
     barrier_sync()
     t0 = time.monotonic()
     if pipeline:
         pipelined_steps()
+    elif pipeline_fan:
+        pipelined_fanout_steps()
     else:
         for k in range(args.steps):
             one_step(k)
@@ -385,7 +444,8 @@ def main():
             # (world × blob_gib, each digest-verified at its destination)
             moved_gib = (1.0 + world) * args.blob_gib * args.steps
             model = f"config4-{args.blob_gib:g}GiB-index-sharded{world}"
-            par = f"fanout{world}-rccl-xgmi-broadcast"
+            par = (f"fanout{world}-rccl-xgmi-broadcast"
+                   + ("-pipelined" if pipeline_fan else ""))
         else:
             moved_gib = 2.0 * args.blob_gib * args.steps * world  # push+pull, all ranks
             model = f"synthetic-{args.blob_gib:g}GiB-blob"

@@ -39,7 +39,7 @@ def test_torchrun_cpu_dry_run_fanout_default(tmp_path, world):
     assert out["scaling"] == "weak"
     assert out["value"] > 0
     assert out["metric"].startswith("push+pull GiB/s")
-    assert out["config"]["parallelism"] == f"fanout{world}-rccl-xgmi-broadcast"
+    assert out["config"]["parallelism"] == f"fanout{world}-rccl-xgmi-broadcast-pipelined"
     assert out["config"]["model"].startswith("config4-")
 
 
