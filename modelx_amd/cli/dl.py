@@ -49,8 +49,13 @@ def main(argv=None) -> int:
     if args.gpus:
         devices = [int(x) for x in args.gpus.split(",") if x != ""]
         from ..client.fanout import fanout_pull_single_process
+        from ..wire import types as wt
 
-        tensors = fanout_pull_single_process(ref, manifest, selection, devices)
+        # leaves sidecars are pull metadata, not model content — don't
+        # land them as HBM tensors
+        gpu_sel = [b for b in selection
+                   if b.media_type != wt.MEDIA_TYPE_MODEL_LEAVES]
+        tensors = fanout_pull_single_process(ref, manifest, gpu_sel, devices)
         for dev, named in tensors.items():
             for name, t in named.items():
                 print(f"cuda:{dev} {name}: {t.numel()} bytes in HBM")
