@@ -8,6 +8,7 @@
 #include <vector>
 
 #include "modelx/http.hpp"
+#include "modelx/json.hpp"
 #include "modelx/store.hpp"
 
 namespace modelx {
@@ -85,6 +86,8 @@ class Registry {
                 const std::string& digest);
   void blob_location(http::Request&, http::ResponseWriter&, const std::string& name,
                      const std::string& digest, const std::string& purpose);
+  void pull_plans(http::Request&, http::ResponseWriter&, const std::string& name);
+  bool build_pull_plan(const std::string& name, const std::string& ref, json::Value* out);
   void pull_plan(http::Request&, http::ResponseWriter&, const std::string& name,
                  const std::string& ref);
   void garbage_collect(http::Request&, http::ResponseWriter&, const std::string& name);

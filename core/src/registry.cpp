@@ -143,6 +143,10 @@ void Registry::handle(http::Request& req, http::ResponseWriter& w) {
     garbage_collect(req, w, name);
     return;
   }
+  if (seg.size() == 3 && seg[2] == "pull-plans" && m == "POST") {
+    pull_plans(req, w, name);
+    return;
+  }
   if (seg.size() == 3 && seg[2] == "index") {
     if (m == "GET") return get_index(req, w, name);
     if (m == "DELETE") return delete_index(req, w, name);
@@ -391,13 +395,52 @@ static std::string base64_encode(const std::string& in) {
 // verify it against the annotation digest, so it is not trusted).
 void Registry::pull_plan(http::Request& req, http::ResponseWriter& w, const std::string& name,
                          const std::string& ref) {
-  constexpr int64_t kInlineLeavesMax = 256 << 10;  // blobs <= 1 GiB at 128 KiB chunks
-  wire::Manifest manifest;
-  if (!store_->GetManifest(name, ref, &manifest)) {
+  json::Value plan;
+  if (!build_pull_plan(name, ref, &plan)) {
     response_error(w,
                    wire::ErrorInfo{404, "MANIFEST_UNKNOWN", "manifest: " + ref + " not found", ""});
     return;
   }
+  response_ok(w, plan);
+}
+
+// POST /{name}/pull-plans  body {"refs": [...]} → {"plans": {ref: plan}}.
+// One control-plane round trip for MANY versions — the per-version GET
+// still left config-5-shaped indexes (one small blob per version, hundreds
+// of versions) paying one round trip per version. Unknown refs are simply
+// omitted; clients fall back per-ref.
+void Registry::pull_plans(http::Request& req, http::ResponseWriter& w,
+                          const std::string& name) {
+  constexpr size_t kMaxRefs = 4096;
+  std::string body = req.read_body_all(kMaxManifestBytes);
+  json::Object plans;
+  try {
+    auto doc = json::parse(body);
+    const auto& refs = doc["refs"].items();
+    if (refs.size() > kMaxRefs) {
+      response_error(w, wire::ErrorInfo{400, "UNKNOWN", "too many refs", ""});
+      return;
+    }
+    for (const auto& r : refs) {
+      const std::string& ref = r.as_string();
+      if (ref.empty() || plans.contains(ref)) continue;
+      json::Value plan;
+      if (build_pull_plan(name, ref, &plan)) plans[ref] = std::move(plan);
+    }
+  } catch (const std::exception& e) {
+    response_error(w, wire::ErrorInfo{400, "UNKNOWN", std::string("bad body: ") + e.what(), ""});
+    return;
+  }
+  json::Object o;
+  o["plans"] = json::Value(std::move(plans));
+  response_ok(w, json::Value(std::move(o)));
+}
+
+bool Registry::build_pull_plan(const std::string& name, const std::string& ref,
+                               json::Value* out) {
+  constexpr int64_t kInlineLeavesMax = 256 << 10;  // blobs <= 1 GiB at 128 KiB chunks
+  wire::Manifest manifest;
+  if (!store_->GetManifest(name, ref, &manifest)) return false;
   json::Object blobs;
   std::vector<const wire::Descriptor*> descs;
   descs.push_back(&manifest.config);
@@ -440,7 +483,8 @@ void Registry::pull_plan(http::Request& req, http::ResponseWriter& w, const std:
   json::Object o;
   o["manifest"] = manifest.to_json();
   o["blobs"] = json::Value(std::move(blobs));
-  response_ok(w, json::Value(std::move(o)));
+  *out = json::Value(std::move(o));
+  return true;
 }
 
 void Registry::garbage_collect(http::Request& req, http::ResponseWriter& w,

@@ -655,8 +655,14 @@ class GpuClient:
         per-version batching alone would never engage)."""
         from concurrent.futures import ThreadPoolExecutor
 
+        # one POST for all versions' plans when the server supports it
+        try:
+            batch = self.remote.get_pull_plans(repository, versions)
+        except Exception:
+            batch = None
+
         def info(v):
-            plan = self.remote.get_pull_plan(repository, v)
+            plan = (batch or {}).get(v) or self.remote.get_pull_plan(repository, v)
             if plan and plan.get("manifest"):
                 return v, types.Manifest.from_dict(plan["manifest"]), plan.get("blobs") or {}
             return v, self.remote.get_manifest(repository, v), {}

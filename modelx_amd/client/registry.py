@@ -182,6 +182,22 @@ class RegistryClient:
         self._raise_for(r)
         return r.json()
 
+    def get_pull_plans(self, repository: str, refs) -> Optional[Dict[str, Any]]:
+        """Batched pull plans: ONE POST for many versions (config-5-shaped
+        indexes store one small blob per version, so the per-version GET
+        still paid a round trip each). Returns {ref: plan} or None when the
+        server lacks the endpoint; unknown refs are omitted (the caller
+        falls back per-ref)."""
+        refs = list(refs)
+        if not refs:
+            return {}
+        r = self.session.post(self._url(repository, "pull-plans"),
+                              headers=self._headers(), json={"refs": refs})
+        if r.status_code in (404, 405, 501):
+            return None
+        self._raise_for(r)
+        return (r.json() or {}).get("plans") or {}
+
     def garbage_collect(self, repository: str) -> Dict[str, Any]:
         r = self.session.post(self._url(repository, "garbage-collect"), headers=self._headers())
         self._raise_for(r)

@@ -234,3 +234,34 @@ def test_pull_plan_endpoint(stack, tmp_path):
 
     # unknown manifest -> None (client falls back and 404s properly)
     assert c.remote.get_pull_plan("plan/model", "nope") is None
+
+
+def test_pull_plans_batch_endpoint(stack, tmp_path):
+    """POST /{name}/pull-plans: ONE round trip for many versions' plans
+    (config-5-shaped indexes store one blob per version). Unknown refs are
+    omitted; each returned plan matches the per-version GET."""
+    mdx, _ = stack
+    c = Client(mdx.url)
+    d = tmp_path / "bp"
+    d.mkdir()
+    from modelx_amd.config import ModelConfig
+
+    (d / "modelx.yaml").write_text(ModelConfig(description="bp").to_yaml())
+    for v in ("v1", "v2", "v3"):
+        (d / "w.bin").write_bytes(os.urandom(64 * 1024) + v.encode())
+        c.push("plan/batch", v, str(d), quiet=True)
+
+    plans = c.remote.get_pull_plans("plan/batch", ["v1", "v2", "v3", "missing"])
+    assert plans is not None
+    assert set(plans) == {"v1", "v2", "v3"}
+    for v in ("v1", "v2", "v3"):
+        single = c.remote.get_pull_plan("plan/batch", v)
+        assert plans[v]["manifest"] == single["manifest"]
+        assert set(plans[v]["blobs"]) == set(single["blobs"])
+    assert c.remote.get_pull_plans("plan/batch", []) == {}
+    # oversized ref lists are rejected, not served
+    import requests
+
+    r = requests.post(mdx.url + "/plan/batch/pull-plans",
+                      json={"refs": ["x"] * 5000}, timeout=10)
+    assert r.status_code == 400
