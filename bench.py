@@ -35,7 +35,12 @@ def parse_args():
     p.add_argument("--blob-gib", type=float, default=8.0,
                    help="per-rank blob size per step (GiB; 8 amortizes per-blob "
                         "control plane, closest bench shape to BASELINE config 2)")
-    p.add_argument("--conns", type=int, default=16, help="ranged-GET connections per rank")
+    p.add_argument("--conns", type=int, default=8,
+                   help="ranged-GET connections per rank (8 measured best under the "
+                        "pipelined step: pull sockets contend with the concurrent "
+                        "push — 23.1-25.0 vs 19.2-21.2 GiB/s at 16, 3x alternating "
+                        "A/B on one box; pull-only workloads still prefer 16, "
+                        "GpuClient's default)")
     p.add_argument("--slot-mib", type=int, default=64)
     p.add_argument("--slots", type=int, default=16)
     p.add_argument("--part-mib", type=int, default=256, help="push part size")
