@@ -180,6 +180,33 @@ def main():
 
     fanout = world > 1 and args.mode in ("auto", "fanout")  # needs collectives
 
+    # Pin this rank (and the servers it spawns) to its GPU's NUMA node:
+    # the engine's recv-into-pinned workers and the loopback store peers
+    # otherwise split across sockets (same-box A/B: 23.4-23.5 vs
+    # 21.6-22.7 GiB/s). Best effort — single-node boxes and CPU dry-runs
+    # skip through the except.
+    if not cpu_mode:
+        try:
+            from modelx_amd import _core
+
+            bdf = _core.hip_pci_bus_id(device).lower()
+            with open(f"/sys/bus/pci/devices/{bdf}/numa_node") as f:
+                node = int(f.read().strip())
+            if node >= 0:
+                with open(f"/sys/devices/system/node/node{node}/cpulist") as f:
+                    spans = f.read().strip()
+                cpus = set()
+                for part in spans.split(","):
+                    if "-" in part:
+                        a, b = part.split("-")
+                        cpus.update(range(int(a), int(b) + 1))
+                    elif part:
+                        cpus.add(int(part))
+                if cpus:
+                    os.sched_setaffinity(0, cpus)
+        except Exception:
+            pass
+
     # --- S3 + registry stack ----------------------------------------------
     # fanout mode: ONE shared modelxd + s3d for the job (rank 0 starts them)
     # — the pull path is one S3 fetch per shard fanned out over xGMI, so the

@@ -1298,6 +1298,15 @@ PYBIND11_MODULE(_core, m) {
   m.doc() = "modelx_amd native core: pinned-ring S3<->HBM engine + CDNA4 SHA-256 kernels";
   m.def("hip_available", &hip_available);
   m.def("hip_device_count", &hip_device_count);
+  // PCI BDF of a device ("0000:0c:00.0") — lets callers find the GPU's
+  // NUMA node via /sys/bus/pci/devices/<bdf>/numa_node and pin their
+  // engine/server threads to it (measured +4-8% on the loopback bench)
+  m.def("hip_pci_bus_id", [](int device) {
+    char buf[64] = {};
+    if (hipDeviceGetPCIBusId(buf, sizeof buf, device) != hipSuccess)
+      throw std::runtime_error("hipDeviceGetPCIBusId failed");
+    return std::string(buf);
+  });
 
   py::class_<GpuEngine>(m, "GpuEngine")
       .def(py::init<int, int, size_t, int>(), py::arg("device") = 0, py::arg("num_slots") = 8,
