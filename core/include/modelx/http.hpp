@@ -133,9 +133,14 @@ struct ClientResponse {
 // One keep-alive connection to host:port (plain TCP; control-plane use).
 class ClientConn {
  public:
-  ClientConn(std::string host, int port) : host_(std::move(host)), port_(port) {}
+  // tls: speak TLS (https presigned URLs / TLS object stores). Certificate
+  // verification uses the system CA paths; MODELX_TLS_INSECURE=1 skips it
+  // (self-signed MinIO/s3d, mirrors the CLI --insecure).
+  ClientConn(std::string host, int port, bool tls = false)
+      : host_(std::move(host)), port_(port), tls_(tls) {}
   ~ClientConn() { close_fd(); }
   bool connected() const { return fd_ >= 0; }
+  bool tls() const { return tls_; }
 
   // Buffered full-body request. body may be empty. Returns false on socket error.
   bool do_request(const std::string& method, const std::string& target, const Headers& headers,
@@ -159,9 +164,14 @@ class ClientConn {
  private:
   bool ensure_connected();
   bool write_full(const char* data, size_t n);
+  bool read_line(std::string* line);
+  // one recv: plain socket or SSL; waitall loops until n bytes (or error)
+  ssize_t conn_recv(char* buf, size_t n, bool waitall);
 
   std::string host_;
   int port_;
+  bool tls_ = false;
+  void* ssl_ = nullptr;  // SSL* when tls_ and connected
   int fd_ = -1;
   std::string rbuf_;
   size_t rpos_ = 0;

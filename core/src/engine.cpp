@@ -146,20 +146,21 @@ struct Range {
 // per-blob latency term that dominates many-small-blob pulls (config 5).
 class ConnPool {
  public:
-  std::unique_ptr<http::ClientConn> checkout(const std::string& host, int port) {
+  std::unique_ptr<http::ClientConn> checkout(const std::string& host, int port, bool tls) {
     std::lock_guard<std::mutex> lk(mu_);
-    auto& v = pool_[host + ":" + std::to_string(port)];
+    auto& v = pool_[(tls ? "https:" : "http:") + host + ":" + std::to_string(port)];
     if (!v.empty()) {
       auto c = std::move(v.back());
       v.pop_back();
       return c;
     }
-    return std::make_unique<http::ClientConn>(host, port);
+    return std::make_unique<http::ClientConn>(host, port, tls);
   }
   void checkin(std::unique_ptr<http::ClientConn> c) {
     if (!c || !c->connected()) return;  // drop broken conns
     std::lock_guard<std::mutex> lk(mu_);
-    auto& v = pool_[c->host() + ":" + std::to_string(c->port())];
+    auto& v = pool_[(c->tls() ? "https:" : "http:") + c->host() + ":" +
+                    std::to_string(c->port())];
     if (v.size() < 64) v.push_back(std::move(c));
   }
 
@@ -313,7 +314,7 @@ class GpuEngine {
     for (int w = 0; w < num_conns; w++) {
       workers.emplace_back([&, w] {
         HIP_CHECK(hipSetDevice(device_));
-        auto conn_holder = conn_pool_.checkout(u.host, u.port);
+        auto conn_holder = conn_pool_.checkout(u.host, u.port, u.scheme == "https");
         http::ClientConn& conn = *conn_holder;
         http::Headers h;
         for (auto& kv : headers) h[kv.first] = kv.second;
@@ -571,7 +572,7 @@ class GpuEngine {
     HIP_CHECK(hipSetDevice(device_));
     double t0 = now_s();
     http::Url u = http::Url::parse(url);
-    auto conn_holder = conn_pool_.checkout(u.host, u.port);
+    auto conn_holder = conn_pool_.checkout(u.host, u.port, u.scheme == "https");
     http::ClientConn& conn = *conn_holder;
     http::Headers h;
     for (auto& kv : headers) h[kv.first] = kv.second;
@@ -1360,6 +1361,22 @@ PYBIND11_MODULE(_core, m) {
         return py::bytes(reinterpret_cast<const char*>(out.data()), out.size());
       },
       py::arg("data"), py::arg("frame_raw") = (uint32_t)(128 << 10));
+  // One-shot GET through the NATIVE http client (the exact code path the
+  // engine's ranged fetches use, incl. TLS) — lets the GPU-less suite prove
+  // https presigned-URL support against a TLS server.
+  m.def("http_get", [](const std::string& url,
+                       const std::map<std::string, std::string>& headers) {
+    http::Url u = http::Url::parse(url);
+    http::ClientConn conn(u.host, u.port, u.scheme == "https");
+    http::Headers h;
+    for (auto& kv : headers) h[kv.first] = kv.second;
+    h["Host"] = u.host + ":" + std::to_string(u.port);
+    http::ClientResponse resp;
+    if (!conn.do_request("GET", u.target(), h, "", &resp))
+      throw std::runtime_error("http_get: request failed (connect/TLS/socket)");
+    return py::make_tuple(resp.status, py::bytes(resp.body));
+  }, py::arg("url"), py::arg("headers") = std::map<std::string, std::string>{});
+
   // CPU-testable canonical sha256 over host bytes — same OpenSSL EVP code
   // the D2H canonical path runs (lets the GPU-less suite oracle it against
   // hashlib)

@@ -521,6 +521,21 @@ void Server::serve_conn(int fd, std::string peer) {
 
 // --------------------------------------------------------------- client ----
 
+// Shared client-side TLS context: system CA paths; MODELX_TLS_INSECURE=1
+// disables verification (self-signed stores).
+static SSL_CTX* client_tls_ctx() {
+  static SSL_CTX* ctx = [] {
+    SSL_CTX* c = SSL_CTX_new(TLS_client_method());
+    if (!c) return (SSL_CTX*)nullptr;
+    SSL_CTX_set_default_verify_paths(c);
+    const char* insecure = getenv("MODELX_TLS_INSECURE");
+    SSL_CTX_set_verify(c, (insecure && *insecure == '1') ? SSL_VERIFY_NONE : SSL_VERIFY_PEER,
+                       nullptr);
+    return c;
+  }();
+  return ctx;
+}
+
 bool ClientConn::ensure_connected() {
   if (fd_ >= 0) return true;
   addrinfo hints{};
@@ -548,10 +563,37 @@ bool ClientConn::ensure_connected() {
   fd_ = fd;
   rbuf_.clear();
   rpos_ = 0;
+  if (tls_) {
+    SSL_CTX* ctx = client_tls_ctx();
+    SSL* ssl = ctx ? SSL_new(ctx) : nullptr;
+    if (!ssl) {
+      close_fd();
+      return false;
+    }
+    SSL_set_fd(ssl, fd_);
+    SSL_set_tlsext_host_name(ssl, host_.c_str());  // SNI
+    const char* insecure = getenv("MODELX_TLS_INSECURE");
+    if (!(insecure && *insecure == '1')) SSL_set1_host(ssl, host_.c_str());
+    int rc;
+    do {
+      rc = SSL_connect(ssl);
+    } while (rc <= 0 && SSL_get_error(ssl, rc) == SSL_ERROR_WANT_READ);
+    if (rc != 1) {
+      SSL_free(ssl);
+      close_fd();
+      return false;
+    }
+    ssl_ = ssl;
+  }
   return true;
 }
 
 void ClientConn::close_fd() {
+  if (ssl_) {
+    SSL_shutdown(static_cast<SSL*>(ssl_));
+    SSL_free(static_cast<SSL*>(ssl_));
+    ssl_ = nullptr;
+  }
   if (fd_ >= 0) {
     ::close(fd_);
     fd_ = -1;
@@ -560,7 +602,44 @@ void ClientConn::close_fd() {
   rpos_ = 0;
 }
 
+ssize_t ClientConn::conn_recv(char* buf, size_t n, bool waitall) {
+  if (ssl_) {
+    SSL* ssl = static_cast<SSL*>(ssl_);
+    size_t got = 0;
+    while (got < n) {
+      int r = SSL_read(ssl, buf + got, static_cast<int>(std::min<size_t>(n - got, 1u << 30)));
+      if (r <= 0) {
+        int err = SSL_get_error(ssl, r);
+        if (err == SSL_ERROR_WANT_READ || err == SSL_ERROR_WANT_WRITE) continue;
+        return got ? static_cast<ssize_t>(got) : (err == SSL_ERROR_ZERO_RETURN ? 0 : -1);
+      }
+      got += static_cast<size_t>(r);
+      if (!waitall) break;
+    }
+    return static_cast<ssize_t>(got);
+  }
+  ssize_t r;
+  do {
+    r = ::recv(fd_, buf, n, waitall ? MSG_WAITALL : 0);
+  } while (r < 0 && errno == EINTR);
+  return r;
+}
+
 bool ClientConn::write_full(const char* data, size_t n) {
+  if (ssl_) {
+    SSL* ssl = static_cast<SSL*>(ssl_);
+    while (n > 0) {
+      int w = SSL_write(ssl, data, static_cast<int>(std::min<size_t>(n, 1u << 30)));
+      if (w <= 0) {
+        int err = SSL_get_error(ssl, w);
+        if (err == SSL_ERROR_WANT_WRITE || err == SSL_ERROR_WANT_READ) continue;
+        return false;
+      }
+      data += w;
+      n -= static_cast<size_t>(w);
+    }
+    return true;
+  }
   while (n > 0) {
     ssize_t w = ::send(fd_, data, n, MSG_NOSIGNAL);
     if (w < 0) {
@@ -598,20 +677,17 @@ bool ClientConn::send_body(const char* data, size_t n) {
   return true;
 }
 
-static bool conn_read_line(int fd, std::string& rbuf, size_t& rpos, std::string* line) {
+bool ClientConn::read_line(std::string* line) {
   line->clear();
   while (true) {
-    if (rpos >= rbuf.size()) {
+    if (rpos_ >= rbuf_.size()) {
       char buf[16384];
-      ssize_t r;
-      do {
-        r = ::recv(fd, buf, sizeof buf, 0);
-      } while (r < 0 && errno == EINTR);
+      ssize_t r = conn_recv(buf, sizeof buf, false);
       if (r <= 0) return false;
-      rbuf.assign(buf, static_cast<size_t>(r));
-      rpos = 0;
+      rbuf_.assign(buf, static_cast<size_t>(r));
+      rpos_ = 0;
     }
-    char c = rbuf[rpos++];
+    char c = rbuf_[rpos_++];
     if (c == '\n') {
       if (!line->empty() && line->back() == '\r') line->pop_back();
       return true;
@@ -622,7 +698,7 @@ static bool conn_read_line(int fd, std::string& rbuf, size_t& rpos, std::string*
 
 bool ClientConn::read_response_head(int* status, Headers* headers) {
   std::string line;
-  if (!conn_read_line(fd_, rbuf_, rpos_, &line)) {
+  if (!read_line(&line)) {
     close_fd();
     return false;
   }
@@ -635,7 +711,7 @@ bool ClientConn::read_response_head(int* status, Headers* headers) {
   keep_alive_ = line.compare(5, 3, "1.1") == 0;
   headers->clear();
   while (true) {
-    if (!conn_read_line(fd_, rbuf_, rpos_, &line)) {
+    if (!read_line(&line)) {
       close_fd();
       return false;
     }
@@ -676,9 +752,7 @@ ssize_t ClientConn::read_body(char* buf, size_t n) {
     rpos_ += take;
     r = static_cast<ssize_t>(take);
   } else {
-    do {
-      r = ::recv(fd_, buf, want, 0);
-    } while (r < 0 && errno == EINTR);
+    r = conn_recv(buf, want, false);
     if (r < 0) {
       close_fd();
       return -1;
@@ -713,11 +787,8 @@ bool ClientConn::read_body_exact(char* buf, size_t n) {
     if (body_remaining_ >= 0 && static_cast<int64_t>(want) > body_remaining_)
       want = static_cast<size_t>(body_remaining_);
     if (want == 0) return false;
-    ssize_t r;
-    do {
-      r = ::recv(fd_, buf + got, want, MSG_WAITALL);
-    } while (r < 0 && errno == EINTR);
-    if (r <= 0) {
+    ssize_t r = conn_recv(buf + got, want, true);
+    if (r <= 0) {  // partial returns retry via the outer loop
       close_fd();
       return false;
     }

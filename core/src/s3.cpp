@@ -115,7 +115,7 @@ http::ClientResponse S3FSProvider::call(const std::string& method, const std::st
     if (!kv.second.empty()) qs += "=" + kv.second;
   }
   if (!qs.empty()) target += "?" + qs;
-  http::ClientConn conn(endpoint_.host, endpoint_.port);
+  http::ClientConn conn(endpoint_.host, endpoint_.port, endpoint_.scheme == "https");
   http::ClientResponse resp;
   if (!conn.do_request(method, target, headers, body, &resp))
     throw std::runtime_error("s3 call failed: " + method + " " + target);

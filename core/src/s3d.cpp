@@ -685,6 +685,7 @@ void handle(http::Request& req, http::ResponseWriter& w) {
 
 int main(int argc, char** argv) {
   std::string listen = ":9000";
+  http::TlsConfig tls;
   for (int i = 1; i < argc; i++) {
     std::string a = argv[i];
     auto next = [&]() -> std::string { return i + 1 < argc ? argv[++i] : ""; };
@@ -694,17 +695,19 @@ int main(int argc, char** argv) {
     else if (a == "--secret-key") g_cfg.secret_key = next();
     else if (a == "--region") g_cfg.region = next();
     else if (a == "--no-auth") g_cfg.verify_auth = false;
+    else if (a == "--tls-cert") tls.cert_file = next();
+    else if (a == "--tls-key") tls.key_file = next();
     else if (a == "--help" || a == "-h") {
       printf("modelx-s3d: S3-compatible test/bench object server\n"
              "  --listen :9000  --root data/s3  --access-key K --secret-key S\n"
-             "  --region us-east-1  --no-auth\n");
+             "  --region us-east-1  --no-auth  --tls-cert F --tls-key F\n");
       return 0;
     }
   }
   signal(SIGPIPE, SIG_IGN);
   mkdirs_for(g_cfg.root + "/.");
   std::thread(trash_collector).detach();
-  http::Server server(listen, handle);
+  http::Server server(listen, handle, tls);
   int port = server.start();
   printf("modelx-s3d listening on port %d root=%s\n", port, g_cfg.root.c_str());
   fflush(stdout);

@@ -109,6 +109,14 @@ class S3Extension(Extension):
             adapter = requests.adapters.HTTPAdapter(pool_connections=32, pool_maxsize=32)
             s.mount("http://", adapter)
             s.mount("https://", adapter)
+            # self-signed TLS object stores: same switch the native engine
+            # honors for https presigned URLs
+            if os.environ.get("MODELX_TLS_INSECURE") == "1":
+                s.verify = False
+                s.trust_env = False  # CURL_CA_BUNDLE would override verify
+                import urllib3
+
+                urllib3.disable_warnings()
             self._local.session = s
         return s
 

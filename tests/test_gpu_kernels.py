@@ -584,3 +584,65 @@ class TestFseWeightsDevice:
         out = ctypes.create_string_buffer(len(data))
         k = z.ZSTD_decompress(out, len(data), blob, len(blob))
         assert not z.ZSTD_isError(k) and out.raw[:k] == data
+
+
+class TestTlsDataPlane:
+    def test_gpu_pull_push_via_https_presigned(self, tmp_path, monkeypatch):
+        """The pinned-ring engine speaks TLS: ranged GETs and multipart
+        part PUTs against https presigned URLs from a TLS object store
+        (production S3 endpoints are TLS; a plain-TCP-only engine would
+        strand the GPU data plane)."""
+        import subprocess
+        import time
+
+        from util_servers import (ACCESS_KEY, BUCKET, MODELXD, S3D, SECRET_KEY,
+                                  ServerProc, _build_servers, free_port, wait_http)
+
+        from modelx_amd import _core
+        from modelx_amd.client.gpu import GpuClient
+
+        _build_servers()
+        d = tmp_path / "certs"
+        d.mkdir()
+        cert, key = str(d / "cert.pem"), str(d / "key.pem")
+        r = subprocess.run(["openssl", "req", "-x509", "-newkey", "rsa:2048",
+                            "-keyout", key, "-out", cert, "-days", "1", "-nodes",
+                            "-subj", "/CN=127.0.0.1"], capture_output=True)
+        if r.returncode != 0:
+            pytest.skip("openssl cert generation failed")
+        monkeypatch.setenv("MODELX_TLS_INSECURE", "1")
+        s3_port = free_port()
+        s3 = ServerProc([S3D, "--listen", f"127.0.0.1:{s3_port}", "--root",
+                         str(tmp_path / "s3"), "--access-key", ACCESS_KEY,
+                         "--secret-key", SECRET_KEY, "--tls-cert", cert,
+                         "--tls-key", key], s3_port)
+        os.makedirs(tmp_path / "s3" / BUCKET, exist_ok=True)
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            try:
+                if _core.http_get(f"https://127.0.0.1:{s3_port}/healthz")[0] == 200:
+                    break
+            except RuntimeError:
+                time.sleep(0.1)
+        mdx_port = free_port()
+        mdx = ServerProc([MODELXD, "--listen", f"127.0.0.1:{mdx_port}", "--s3-url",
+                          f"https://127.0.0.1:{s3_port}", "--s3-bucket", BUCKET,
+                          "--s3-access-key", ACCESS_KEY, "--s3-secret-key",
+                          SECRET_KEY, "--enable-redirect"], mdx_port)
+        wait_http(mdx_port)
+        try:
+            g = GpuClient(f"http://127.0.0.1:{mdx_port}", device=0,
+                          num_slots=4, slot_bytes=8 << 20)
+            src = torch.randint(0, 256, ((20 << 20) + 77,), dtype=torch.uint8,
+                                device="cuda:0")
+            # push parts go D2H -> https PUT; pull is https ranged GETs -> HBM
+            g.push_from_gpu("tls/gpu", "v1", {"w.bin": src}, part_bytes=4 << 20)
+            desc = next(b for b in g.remote.get_manifest("tls/gpu", "v1").blobs
+                        if b.name == "w.bin")
+            url, _ = g._download_url("tls/gpu", desc)
+            assert url.startswith("https://")
+            back = g.pull_to_gpu("tls/gpu", "v1")
+            assert torch.equal(back["w.bin"], src)
+        finally:
+            mdx.stop()
+            s3.stop()
