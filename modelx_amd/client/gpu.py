@@ -50,6 +50,19 @@ def _core():
     return core
 
 
+def _tls_verify() -> bool:
+    """requests verify flag for presigned-URL calls; MODELX_TLS_INSECURE=1
+    (self-signed stores) matches the native engine's switch."""
+    import os
+
+    if os.environ.get("MODELX_TLS_INSECURE") == "1":
+        import urllib3
+
+        urllib3.disable_warnings()
+        return False
+    return True
+
+
 def _signed_headers(part: dict) -> Dict[str, str]:
     out = {}
     for k, v in (part.get("signedHeader") or {}).items():
@@ -723,7 +736,8 @@ class GpuClient:
             import requests
 
             url, headers = located
-            resp = requests.get(url, headers=headers, stream=True)
+            resp = requests.get(url, headers=headers, stream=True,
+                                verify=_tls_verify())
             resp.raise_for_status()
             source = resp.iter_content(chunk_size=1 << 20)
         else:
@@ -843,8 +857,8 @@ class GpuClient:
             for attempt in range(self.PART_RETRIES):
                 try:
                     requests.request(p.get("method") or "PUT", p["url"],
-                                     headers=_signed_headers(p),
-                                     data=data).raise_for_status()
+                                     headers=_signed_headers(p), data=data,
+                                     verify=_tls_verify()).raise_for_status()
                     return
                 except Exception:
                     if attempt == self.PART_RETRIES - 1:
