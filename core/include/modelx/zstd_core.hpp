@@ -467,25 +467,61 @@ MX_HD static inline int huf_decode_stream(const HufTable* t, const u8* src, u64 
   if (!bb.init(src, srclen)) return dstlen == 0 ? MXZ_OK : MXZ_ERR_HUFFMAN;
   u64 produced = 0;
   u32 mb = t->maxbits;
-  // fast path: local cursor arithmetic (same semantics as bb.read)
+  i64 bits = bb.bits;
+  const u8* sp = bb.src;
+  const u32 mask = (1u << mb) - 1;
+  // fast path: one unaligned 64-bit window load serves many symbols — the
+  // per-symbol 9-byte bounded gather below was the measured serial tail of
+  // literal-heavy (bf16 tensor) frames. The window is anchored at bit `lo`
+  // (LSB), holding bits [lo, lo+64); symbols read the top `mb` bits below
+  // the frontier. Requires 8 loadable bytes at byte(lo) — the last few
+  // symbols fall through to the bounded loop.
   while (produced < dstlen) {
-    i64 lo = bb.bits - (i64)mb;
+    i64 lo = bits - 57;
+    if (lo < 0) break;  // near the stream start: bounded path
+    u64 byte0 = (u64)(lo >> 3);
+    if (byte0 + 8 > srclen) {
+      // shift the window down so the 8-byte load stays inside the stream
+      byte0 = srclen - 8;  // srclen >= 8 guaranteed: bits >= 57 here
+      lo = (i64)(byte0 << 3);
+    }
+    u64 v;
+    __builtin_memcpy(&v, sp + byte0, 8);
+    v >>= (lo & 7);
+    u32 have = (u32)(bits - lo);  // <= 57 + 7
+    if (have > 57) {              // keep idx math in-range after the shift-down
+      v >>= (have - 57);
+      lo += have - 57;
+      have = 57;
+    }
+    while (have >= mb && produced < dstlen) {
+      u32 idx = (u32)(v >> (have - mb)) & mask;
+      u32 nb = t->nbits[idx];
+      dst[produced++] = t->symbol[idx];
+      bits -= nb;
+      have -= nb;
+    }
+    if (have >= mb) break;  // dstlen reached
+  }
+  // bounded tail (original semantics, incl. zero-padded final reads)
+  while (produced < dstlen) {
+    i64 lo = bits - (i64)mb;
     u32 idx;
     {
       i64 l = lo < 0 ? 0 : lo;
       u64 byte0 = (u64)(l >> 3);
       u64 v = 0;
-      for (u32 i = 0; i < 9 && byte0 + i <= (u64)((bb.bits - 1) >> 3); i++)
-        v |= (u64)bb.src[byte0 + i] << (8 * i);
+      for (u32 i = 0; i < 9 && byte0 + i <= (u64)((bits - 1) >> 3); i++)
+        v |= (u64)sp[byte0 + i] << (8 * i);
       v >>= (l & 7);
-      u32 avail = (u32)(bb.bits - l);
+      u32 avail = (u32)(bits - l);
       idx = (u32)(v & (((u64)1 << avail) - 1));
       if (lo < 0) idx <<= (u32)(-lo);
     }
     u32 nb = t->nbits[idx];
-    if ((i64)nb > bb.bits) return MXZ_ERR_HUFFMAN;  // ran out of bits
+    if ((i64)nb > bits) return MXZ_ERR_HUFFMAN;  // ran out of bits
     dst[produced++] = t->symbol[idx];
-    bb.bits -= nb;
+    bits -= nb;
   }
   return MXZ_OK;
 }
