@@ -7,6 +7,12 @@ REPO="$(cd "$(dirname "$0")/.." && pwd)"
 export TMPDIR=/tmp
 mkdir -p "$REPO/gpurun_out"
 cd /tmp
+rocprofv3 --pmc must NOT be combined with trace domains (pool rule).
+set -x
+REPO="$(cd "$(dirname "$0")/.." && pwd)"
+export TMPDIR=/tmp
+mkdir -p "$REPO/gpurun_out"
+cd /tmp
 cat > /tmp/pmc_workload.py <<'EOF'
 import torch
 from modelx_amd import _core
@@ -31,7 +37,7 @@ assert m == n and torch.equal(back, data)
 print("workload ok")
 EOF
 rocprofv3 --pmc SQ_INSTS_VALU SQ_LDS_BANK_CONFLICT -d "$REPO/gpurun_out/pmc_r2" -o r2 -- \
-  bash -c "cd '$REPO' && python /tmp/pmc_workload.py > gpurun_out/pmc_workload.log 2>&1"
+  bash -c "cd '$REPO' && python tools/pmc_workload_r2.py > gpurun_out/pmc_workload.log 2>&1"
 rc=$?
 echo "pmc_rc=$rc"
 ls "$REPO/gpurun_out/pmc_r2" 2>/dev/null
