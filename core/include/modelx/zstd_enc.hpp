@@ -453,24 +453,14 @@ MX_HD static inline u32 lz_parse(const u8* base, u64 block_off, u64 block_len, u
 
 // Byte histogram. Device: lane-strided with LDS atomics (single-wave
 // kernel; all lanes see the final counts after the barrier).
-// rep: 4×257 u32 LDS scratch (replicated histograms). Tensor-byte payloads
-// cluster on few symbol values, so 64 lanes atomicAdd-ing ONE address
-// serialize; adjacent lanes hash to different replicas (and the 257 stride
-// decorrelates banks), then the replicas merge. PMC: 586 M LDS bank
-// conflicts in the compress kernel before this.
-MX_HD static inline void lit_histogram(const u8* lit, u64 n, u32* hist, u32* rep) {
+MX_HD static inline void lit_histogram(const u8* lit, u64 n, u32* hist) {
 #if defined(__HIP_DEVICE_COMPILE__)
   u32 lane = mx_lane(), w = mx_width();
-  for (u32 i = lane; i < 4 * 257; i += w) rep[i] = 0;
+  for (u32 i = lane; i < 256; i += w) hist[i] = 0;
   mx_sync();
-  u32* mine = rep + (lane & 3u) * 257;
-  for (u64 i = lane; i < n; i += w) atomicAdd(&mine[lit[i]], 1u);
-  mx_sync();
-  for (u32 i = lane; i < 256; i += w)
-    hist[i] = rep[i] + rep[257 + i] + rep[2 * 257 + i] + rep[3 * 257 + i];
+  for (u64 i = lane; i < n; i += w) atomicAdd(&hist[lit[i]], 1u);
   mx_sync();
 #else
-  (void)rep;
   for (u32 i = 0; i < 256; i++) hist[i] = 0;
   for (u64 i = 0; i < n; i++) hist[lit[i]]++;
 #endif
@@ -735,7 +725,6 @@ struct EncTables {
   FseEnc wfse;  // huffman-weight FSE encoder (full-byte alphabets)
   HufEnc he;
   u32 lit_hist[256];
-  u32 hist_rep[4 * 257];  // replicated-histogram scratch (see lit_histogram)
   u32 scan_tmp[64];  // cross-lane exchange for the parallel huffman encoder
 };
 
@@ -769,7 +758,7 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
   if (nseq == 0) {
     // no matches: RLE block for a single repeated byte, else try a
     // huffman-literals-only compressed block, else raw
-    lit_histogram(block, block_len, et->lit_hist, et->hist_rep);
+    lit_histogram(block, block_len, et->lit_hist);
     u32 nsym = 0;
     for (u32 i = 0; i < 256; i++)
       if (et->lit_hist[i]) nsym++;
@@ -814,7 +803,7 @@ MX_HD static inline i64 encode_block(const u8* src, u64 block_off, u64 block_len
       }
       mx_par_copy(lit_buf + lw, src + p, (block_off + block_len) - p);
     }
-    lit_histogram(lit_buf, lit_total, et->lit_hist, et->hist_rep);
+    lit_histogram(lit_buf, lit_total, et->lit_hist);
     u32 nsym = 0;
     for (u32 i = 0; i < 256; i++)
       if (et->lit_hist[i]) nsym++;
